@@ -1,0 +1,138 @@
+"""dear_pytorch_amd — MI355X-native decoupled-all-reduce (DeAR) DP training engine.
+
+Horovod-shaped public API with capability parity to the reference
+lzhangbv/dear_pytorch (dear/__init__.py + dear/dopt_rsag.py L3/L4 surface):
+
+    import dear_pytorch_amd as dear
+    dear.init()                                   # torchrun / env:// rendezvous
+    opt = dear.DistributedOptimizer(opt, model=model)
+    dear.broadcast_parameters(model.state_dict(), root_rank=0)
+    dear.broadcast_optimizer_state(opt, root_rank=0)
+
+Process model: one process per GPU, ``torch.distributed`` over RCCL/xGMI
+(backend "nccl" IS RCCL on ROCm).  The reference bootstraps with MPI at import
+time (dopt_rsag.py:32); here ``init()`` performs the rendezvous explicitly.
+"""
+from __future__ import annotations
+
+import datetime
+import os
+
+import torch
+import torch.distributed as dist
+
+from .parallel.dear import DistributedOptimizer, DearOptimizer  # noqa: F401
+from .parallel.fusion import build_groups, BucketGroup  # noqa: F401
+from .comm.backend import create_backend, CommBackend  # noqa: F401
+
+__version__ = "0.1.0"
+
+_generic_backend: CommBackend | None = None
+
+
+def init(backend: str | None = None, timeout_s: int = 1800):
+    """Initialize the distributed world from torchrun env vars.
+
+    backend: None picks "nccl" (RCCL) when a GPU is visible, else "gloo".
+    Safe to call when WORLD_SIZE is absent (single-process mode).
+    """
+    global _generic_backend
+    if dist.is_initialized():
+        return
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1 and "MASTER_ADDR" not in os.environ:
+        _generic_backend = create_backend()
+        return
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if torch.cuda.is_available():
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    dist.init_process_group(backend=backend,
+                            timeout=datetime.timedelta(seconds=timeout_s))
+    _generic_backend = None  # created lazily after init
+
+
+def _generic():
+    global _generic_backend
+    if _generic_backend is None:
+        _generic_backend = create_backend("generic")
+    return _generic_backend
+
+
+def is_initialized() -> bool:
+    return dist.is_initialized()
+
+
+def rank() -> int:
+    return dist.get_rank() if dist.is_initialized() else 0
+
+
+def size() -> int:
+    return dist.get_world_size() if dist.is_initialized() else 1
+
+
+def local_rank() -> int:
+    return int(os.environ.get("LOCAL_RANK", "0"))
+
+
+def shutdown():
+    global _generic_backend
+    _generic_backend = None
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+def allreduce(tensor: torch.Tensor, average: bool = True, name: str | None = None):
+    """Blocking mean/sum all-reduce of a (typically small metric) tensor —
+    reference dear/dopt_rsag.py:543."""
+    t = tensor.detach().clone()
+    _generic().all_reduce(t).host_wait()
+    if average and size() > 1:
+        t /= size()
+    return t
+
+
+def broadcast_parameters(params, root_rank: int = 0):
+    """Broadcast model parameters (a state_dict or named-parameter iterable)
+    from root so every rank starts identical — reference dopt_rsag.py:396-421."""
+    if size() == 1:
+        return
+    if isinstance(params, dict):
+        items = sorted(params.items())
+    else:
+        items = sorted(dict(params).items())
+    be = _generic()
+    handles = [be.broadcast(v.data, root_rank) for _, v in items
+               if torch.is_tensor(v)]
+    for h in handles:
+        h.host_wait()
+
+
+def broadcast_optimizer_state(optimizer, root_rank: int = 0):
+    """Broadcast optimizer state (incl. python scalars via tensor wrapping) —
+    reference dopt_rsag.py:424-540."""
+    if size() == 1:
+        return
+    if isinstance(optimizer, DearOptimizer):
+        optimizer = optimizer.optim
+    # one-shot startup consistency: ship root's full state (scalars wrapped
+    # with the tensors, mirroring the reference's tensor-wrap callbacks)
+    obj = [optimizer.state_dict() if rank() == root_rank else None]
+    dist.broadcast_object_list(obj, src=root_rank)
+    if rank() != root_rank:
+        cpu_state = obj[0]
+        dev = next(iter(
+            p.device for g in optimizer.param_groups for p in g["params"]))
+        for st in cpu_state["state"].values():
+            for k, v in st.items():
+                if torch.is_tensor(v) and v.dim() > 0:
+                    st[k] = v.to(dev)
+        optimizer.load_state_dict(cpu_state)
+
+
+# convenience namespace parity with the reference benchmark drivers
+def broadcast_object(obj, root_rank: int = 0):
+    lst = [obj]
+    if size() > 1:
+        dist.broadcast_object_list(lst, src=root_rank)
+    return lst[0]

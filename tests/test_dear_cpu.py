@@ -1,0 +1,143 @@
+"""DeAR-vs-serial numeric equivalence (the reference's implicit correctness
+contract, SURVEY.md §4: decoupled lazy update == SGD up to one-iteration-late
+semantics). CPU, world_size 1 here; world_size 2 in test_dear_dist.py."""
+import copy
+
+import pytest
+import torch
+import torch.nn as nn
+
+import dear_pytorch_amd as dear
+
+
+def _model(seed=0):
+    torch.manual_seed(seed)
+    return nn.Sequential(
+        nn.Linear(16, 32), nn.ReLU(),
+        nn.Linear(32, 32), nn.Tanh(),
+        nn.Linear(32, 4),
+    )
+
+
+def _data(T=6, bs=8, seed=42):
+    g = torch.Generator().manual_seed(seed)
+    return [(torch.randn(bs, 16, generator=g), torch.randn(bs, 4, generator=g))
+            for _ in range(T)]
+
+
+def _train_serial(model, opt_fn, data):
+    opt = opt_fn(model.parameters())
+    for x, y in data:
+        opt.zero_grad()
+        nn.functional.mse_loss(model(x), y).backward()
+        opt.step()
+    return model
+
+
+def _train_dear(model, opt_fn, data, **kw):
+    opt = dear.DistributedOptimizer(opt_fn(model.parameters()), model=model, **kw)
+    for x, y in data:
+        opt.zero_grad()
+        nn.functional.mse_loss(model(x), y).backward()
+        opt.step()
+    opt.synchronize()  # apply the last pending lazy update
+    return model
+
+
+OPTS = {
+    "sgd": lambda ps: torch.optim.SGD(ps, lr=0.05),
+    "sgd_mom": lambda ps: torch.optim.SGD(ps, lr=0.05, momentum=0.9,
+                                          weight_decay=1e-4),
+    "sgd_nesterov": lambda ps: torch.optim.SGD(ps, lr=0.05, momentum=0.9,
+                                               nesterov=True),
+    "sgd_dampening": lambda ps: torch.optim.SGD(ps, lr=0.05, momentum=0.9,
+                                                dampening=0.3),
+    "adam": lambda ps: torch.optim.Adam(ps, lr=1e-3, weight_decay=1e-4),
+    "adamw": lambda ps: torch.optim.AdamW(ps, lr=1e-3, weight_decay=1e-2),
+    "rmsprop": lambda ps: torch.optim.RMSprop(ps, lr=1e-3),  # shadow path
+}
+
+
+@pytest.mark.parametrize("name", list(OPTS))
+@pytest.mark.parametrize("threshold", [None, 1 << 12, 1 << 30])
+def test_dear_matches_serial(name, threshold):
+    data = _data()
+    a = _train_serial(_model(), OPTS[name], data)
+    b = _train_dear(_model(), OPTS[name], data, threshold_bytes=threshold)
+    for (na, pa), (nb, pb) in zip(a.named_parameters(), b.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), \
+            f"{name} thr={threshold}: {na} max diff " \
+            f"{(pa - pb).abs().max().item():.3e}"
+
+
+def test_lazy_update_is_one_iteration_late():
+    data = _data(T=3)
+    m = _model()
+    before = copy.deepcopy(m.state_dict())
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.1), model=m)
+    x, y = data[0]
+    nn.functional.mse_loss(m(x), y).backward()
+    opt.step()
+    # weights unchanged until the NEXT forward touches each group
+    for k, v in m.state_dict().items():
+        assert torch.equal(v, before[k])
+    m(x)  # forward applies the pending update group by group
+    changed = any(not torch.equal(v, before[k])
+                  for k, v in m.state_dict().items())
+    assert changed
+
+
+def test_exclude_parts_flags_run():
+    data = _data(T=3)
+    for part in ("reducescatter", "allgather"):
+        _train_dear(_model(), OPTS["sgd"], data, exclude_parts=part)
+
+
+def test_regroup_between_iterations():
+    data = _data(T=6)
+    a = _train_serial(_model(), OPTS["sgd_mom"], data)
+    m = _model()
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9,
+                        weight_decay=1e-4), model=m)
+    for i, (x, y) in enumerate(data):
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+        if i == 2:
+            opt.regroup(1 << 12)  # BO-tuner style mid-training rebucket
+    opt.synchronize()
+    for (na, pa), (nb, pb) in zip(a.named_parameters(), m.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), na
+
+
+def test_regroup_preserves_momentum_state():
+    m = _model()
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m)
+    data = _data(T=4)
+    for x, y in data[:2]:
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+    opt.synchronize()
+    mom_before = {id(p): opt.optim.state[p]["momentum_buffer"].clone()
+                  for g in opt.groups for s in g.slots for p in [s.param]}
+    opt.regroup(1 << 14)
+    for g in opt.groups:
+        for s in g.slots:
+            assert torch.equal(opt.optim.state[s.param]["momentum_buffer"],
+                               mom_before[id(s.param)])
+
+
+def test_state_dict_roundtrip():
+    m = _model()
+    opt = dear.DistributedOptimizer(
+        torch.optim.Adam(m.parameters(), lr=1e-3), model=m)
+    data = _data(T=3)
+    for x, y in data:
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+    opt.synchronize()
+    sd = opt.state_dict()
+    assert sd["state"], "optimizer state should be populated"
+    opt.load_state_dict(sd)
