@@ -1,0 +1,178 @@
+#!/usr/bin/env python
+"""Flagship benchmark: DeAR data-parallel training throughput on MI355X.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+launched (N>1) via torch.distributed.run with one rank per GPU over RCCL.
+Rank 0 prints ONE JSON line: whole-job samples/sec on the BASELINE.json
+config (ResNet-50 bs64 fake ImageNet, fp32 — the reference's headline
+`Total img/sec on N GPU(s)` protocol, dear/imagenet_benchmark.py:159-172;
+--model bert_large for the BERT-Large seq128 headline).
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--model", default="resnet50",
+                   help="resnet50|vgg16|densenet201|inceptionv4|bert_base|bert_large")
+    p.add_argument("--batch-size", type=int, default=None,
+                   help="per-GPU batch (default: 64 CNN / 32 BERT)")
+    p.add_argument("--seq-len", type=int, default=128)
+    p.add_argument("--method", default="dear",
+                   choices=["dear", "ddp", "wfbp", "mgwfbp", "naive", "rb"],
+                   help="gradient-sync method (dear is the product)")
+    p.add_argument("--threshold-mb", type=float, default=25.0)
+    p.add_argument("--exclude-parts", default="")
+    p.add_argument("--no-fusion", action="store_true")
+    return p.parse_args()
+
+
+def build_workload(args, device):
+    from dear_pytorch_amd import models
+    is_bert = args.model.startswith("bert")
+    if is_bert:
+        bs = args.batch_size or 32
+        cfg = models.bert_large() if args.model == "bert_large" else \
+            models.bert_base()
+        model = models.BertForPreTraining(cfg).to(device)
+        crit = models.BertPretrainingCriterion(cfg.vocab_size).to(device)
+        S = args.seq_len
+        g = torch.Generator().manual_seed(1234)
+        ids = torch.randint(0, cfg.vocab_size, (bs, S), generator=g).to(device)
+        tt = torch.zeros(bs, S, dtype=torch.long, device=device)
+        mask = torch.ones(bs, S, dtype=torch.long, device=device)
+        mlm = torch.full((bs, S), -1, dtype=torch.long)
+        sel = torch.rand(bs, S, generator=g) < 0.15
+        mlm[sel] = torch.randint(0, cfg.vocab_size, (int(sel.sum()),),
+                                 generator=g)
+        mlm = mlm.to(device)
+        nsp = torch.randint(0, 2, (bs,), generator=g).to(device)
+
+        def step_fn(model, opt):
+            opt.zero_grad()
+            scores, seq_rel = model(ids, tt, mask)
+            loss = crit(scores, seq_rel, mlm, nsp)
+            loss.backward()
+            opt.step()
+
+        opt_fn = lambda ps: torch.optim.SGD(ps, lr=2e-5)  # reference :122
+        unit, metric = "sen/sec", "sen/sec"
+    else:
+        bs = args.batch_size or 64
+        model = models.get_cnn(args.model).to(device)
+        g = torch.Generator().manual_seed(1234)
+        res = 299 if args.model == "inceptionv4" else 224
+        data = torch.randn(bs, 3, res, res, generator=g).to(device)
+        target = torch.randint(0, 1000, (bs,), generator=g).to(device)
+        lossf = torch.nn.CrossEntropyLoss().to(device)
+
+        def step_fn(model, opt):
+            opt.zero_grad()
+            loss = lossf(model(data), target)
+            loss.backward()
+            opt.step()
+
+        opt_fn = lambda ps: torch.optim.SGD(ps, lr=0.01, momentum=0.9)
+        unit, metric = "img/sec", "img/sec"
+    return model, opt_fn, step_fn, bs, unit, metric
+
+
+def wrap_method(args, model, opt_fn):
+    import dear_pytorch_amd as dear
+    threshold = None if args.no_fusion else int(args.threshold_mb * 1024 * 1024)
+    if args.method == "ddp":
+        if dear.size() > 1:
+            model = torch.nn.parallel.DistributedDataParallel(
+                model, bucket_cap_mb=args.threshold_mb,
+                gradient_as_bucket_view=True)
+        opt = opt_fn(model.parameters())
+        return model, opt
+    if args.method == "dear":
+        opt = dear.DistributedOptimizer(opt_fn(model.parameters()),
+                                        model=model, threshold_bytes=threshold,
+                                        exclude_parts=args.exclude_parts)
+        return model, opt
+    from dear_pytorch_amd.parallel import baselines
+    opt = baselines.make(args.method, opt_fn(model.parameters()), model,
+                         threshold_bytes=threshold)
+    return model, opt
+
+
+def main():
+    args = parse_args()
+    import dear_pytorch_amd as dear
+    dear.init()
+    rank, world = dear.rank(), dear.size()
+    on_gpu = torch.cuda.is_available()
+    device = torch.device("cuda", dear.local_rank()) if on_gpu else \
+        torch.device("cpu")
+    if on_gpu:
+        torch.cuda.set_device(device)
+        torch.backends.cudnn.benchmark = True
+
+    model, opt_fn, step_fn, bs, unit, metric = build_workload(args, device)
+    if world > 1:
+        dear.broadcast_parameters(model.state_dict(), root_rank=0)
+    model, opt = wrap_method(args, model, opt_fn)
+    model.train()
+
+    def sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step_fn(model, opt)
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step_fn(model, opt)
+    sync()
+    elapsed = time.perf_counter() - t0
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        t = t.to(device) if torch.distributed.get_backend() == "nccl" else t
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    value = world * bs * args.steps / elapsed
+    if rank == 0:
+        out = {
+            "metric": metric,
+            "value": round(value, 2),
+            "unit": unit,
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": world * bs,
+                "seq_len": args.seq_len if args.model.startswith("bert") else None,
+                "parallelism": f"dp{world}",
+                "method": args.method,
+                "threshold_mb": None if args.no_fusion else args.threshold_mb,
+            },
+        }
+        print(json.dumps(out), flush=True)
+    dear.shutdown()
+
+
+if __name__ == "__main__":
+    main()

@@ -270,7 +270,8 @@ class RcclBackend(CommBackend):
 
     def all_gather(self, shard, bucket, after=None):
         if isinstance(after, _EventHandle):
-            self.comm.wait_op_comm(after.comm, after.opid)
+            import dear_pytorch_amd._comm_core as comm_core
+            comm_core.wait_op_across(self.comm, after.comm, after.opid)
         elif after is not None:
             after.wait_compute()
         self.record_compute()

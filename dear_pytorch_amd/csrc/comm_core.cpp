@@ -1,0 +1,316 @@
+// MI355X-native RCCL communicator for the DeAR engine.
+//
+// Capability parity with the reference's common/comm_core/src/communicator.cpp
+// (NCCL on CUDA side streams, MPI bootstrap, integer stream-index handles) —
+// redesigned for RCCL over xGMI:
+//   * bootstrap: the rcclUniqueId arrives from Python (exchanged over the
+//     torch.distributed store) — no MPI dependency;
+//   * one RCCL communicator per Communicator instance, on a dedicated HIP
+//     side stream from the ATen pool (so PyTorch's allocator/stream machinery
+//     knows it);
+//   * hipEvent-based op handles: every collective records an event; waits are
+//     expressed DEVICE-side (hipStreamWaitEvent against the compute stream or
+//     another communicator's stream), fixing the reference's host-blocking
+//     int-index handles and the flag bugs on reduceScatter/allGather
+//     (communicator.cpp:157-183) and its placebo self-stream wait
+//     (tensorfusion.py:304);
+//   * full dtype dispatch (fp32/fp64/fp16/bf16/i32/i64/u8) instead of the
+//     reference's fp32/int64-only paths;
+//   * errors surface as C++ exceptions (→ Python), not exit(1).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+#include <c10/hip/HIPGuard.h>
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <deque>
+#include <map>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#define HIP_CHECK(cmd)                                                        \
+  do {                                                                        \
+    hipError_t e_ = (cmd);                                                    \
+    if (e_ != hipSuccess)                                                     \
+      throw std::runtime_error(std::string("HIP error: ") +                   \
+                               hipGetErrorString(e_) + " at " #cmd);          \
+  } while (0)
+
+#define RCCL_CHECK(cmd)                                                       \
+  do {                                                                        \
+    ncclResult_t r_ = (cmd);                                                  \
+    if (r_ != ncclSuccess)                                                    \
+      throw std::runtime_error(std::string("RCCL error: ") +                  \
+                               ncclGetErrorString(r_) + " at " #cmd);         \
+  } while (0)
+
+namespace {
+
+ncclDataType_t rccl_dtype(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return ncclFloat32;
+    case at::kDouble: return ncclFloat64;
+    case at::kHalf: return ncclFloat16;
+    case at::kBFloat16: return ncclBfloat16;
+    case at::kInt: return ncclInt32;
+    case at::kLong: return ncclInt64;
+    case at::kByte: return ncclUint8;
+    default:
+      throw std::runtime_error("unsupported dtype for RCCL collective: " +
+                               std::string(t.toString()));
+  }
+}
+
+void check_flat(const at::Tensor& t) {
+  TORCH_CHECK(t.is_cuda(), "collective input must be a GPU tensor");
+  TORCH_CHECK(t.is_contiguous(), "collective input must be contiguous");
+}
+
+}  // namespace
+
+class Communicator {
+ public:
+  Communicator(int rank, int size, const std::string& uid_bytes)
+      : rank_(rank), size_(size),
+        device_(c10::hip::current_device()),
+        stream_(c10::hip::getStreamFromPool(/*isHighPriority=*/true, device_)) {
+    TORCH_CHECK(uid_bytes.size() == sizeof(ncclUniqueId),
+                "bad rccl unique id size");
+    ncclUniqueId uid;
+    memcpy(&uid, uid_bytes.data(), sizeof(uid));
+    RCCL_CHECK(ncclCommInitRank(&comm_, size_, uid, rank_));
+  }
+
+  ~Communicator() {
+    if (comm_) ncclCommDestroy(comm_);
+    for (auto& e : event_pool_) hipEventDestroy(e);
+    for (auto& kv : live_events_) hipEventDestroy(kv.second);
+  }
+
+  int rank() const { return rank_; }
+  int size() const { return size_; }
+  uint64_t stream_handle() const { return (uint64_t)stream_.stream(); }
+
+  // ---- ordering edges ----------------------------------------------------
+  // Make the comm stream wait for the given (compute) stream's current work:
+  // the compute→RS edge the reference faked with a self-wait.
+  void wait_stream(uint64_t other_stream) {
+    hipEvent_t ev = acquire_event();
+    HIP_CHECK(hipEventRecord(ev, (hipStream_t)other_stream));
+    HIP_CHECK(hipStreamWaitEvent(stream_.stream(), ev, 0));
+    release_event(ev);
+  }
+
+  // Make a given stream (e.g. the compute stream) wait for op `id`.
+  void wait_op_stream(int64_t id, uint64_t stream) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = live_events_.find(id);
+    if (it == live_events_.end()) return;  // already collected
+    HIP_CHECK(hipStreamWaitEvent((hipStream_t)stream, it->second, 0));
+  }
+
+  // Make THIS communicator's stream wait for op `id` of another communicator
+  // (RS→AG edge across the two side streams).
+  static void wait_op_on(Communicator& self, Communicator& other, int64_t id) {
+    std::lock_guard<std::mutex> g(other.mu_);
+    auto it = other.live_events_.find(id);
+    if (it == other.live_events_.end()) return;
+    HIP_CHECK(hipStreamWaitEvent(self.stream_.stream(), it->second, 0));
+  }
+
+  void wait_op_host(int64_t id) {
+    hipEvent_t ev = nullptr;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      auto it = live_events_.find(id);
+      if (it == live_events_.end()) return;
+      ev = it->second;
+    }
+    HIP_CHECK(hipEventSynchronize(ev));
+  }
+
+  bool op_done(int64_t id) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = live_events_.find(id);
+    if (it == live_events_.end()) return true;
+    return hipEventQuery(it->second) == hipSuccess;
+  }
+
+  void synchronize() { HIP_CHECK(hipStreamSynchronize(stream_.stream())); }
+
+  // ---- collectives (async on the side stream; return event-handle id) ----
+  int64_t all_reduce(at::Tensor t) {
+    check_flat(t);
+    RCCL_CHECK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                             rccl_dtype(t), ncclSum, comm_, stream_.stream()));
+    return record();
+  }
+
+  int64_t reduce(at::Tensor t, int root) {
+    check_flat(t);
+    RCCL_CHECK(ncclReduce(t.data_ptr(), t.data_ptr(), t.numel(), rccl_dtype(t),
+                          ncclSum, root, comm_, stream_.stream()));
+    return record();
+  }
+
+  int64_t broadcast(at::Tensor t, int root) {
+    check_flat(t);
+    RCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
+                             rccl_dtype(t), root, comm_, stream_.stream()));
+    return record();
+  }
+
+  int64_t reduce_scatter(at::Tensor send, at::Tensor recv) {
+    check_flat(send);
+    check_flat(recv);
+    TORCH_CHECK(send.numel() == recv.numel() * size_,
+                "reduce_scatter: send numel must be recv numel * world_size");
+    RCCL_CHECK(ncclReduceScatter(send.data_ptr(), recv.data_ptr(),
+                                 recv.numel(), rccl_dtype(send), ncclSum,
+                                 comm_, stream_.stream()));
+    return record();
+  }
+
+  int64_t all_gather(at::Tensor send, at::Tensor recv) {
+    check_flat(send);
+    check_flat(recv);
+    TORCH_CHECK(recv.numel() == send.numel() * size_,
+                "all_gather: recv numel must be send numel * world_size");
+    RCCL_CHECK(ncclAllGather(send.data_ptr(), recv.data_ptr(), send.numel(),
+                             rccl_dtype(send), comm_, stream_.stream()));
+    return record();
+  }
+
+  // reduce-to-root + broadcast decomposition (reference allReduceRB,
+  // communicator.cpp:185-196) — kept for the dopt_rb ablation.
+  int64_t all_reduce_rb(at::Tensor t, int root) {
+    check_flat(t);
+    RCCL_CHECK(ncclReduce(t.data_ptr(), t.data_ptr(), t.numel(), rccl_dtype(t),
+                          ncclSum, root, comm_, stream_.stream()));
+    RCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
+                             rccl_dtype(t), root, comm_, stream_.stream()));
+    return record();
+  }
+
+  // RS+AG in-place allreduce over a %P-padded tensor (reference allReduceRSAG,
+  // communicator.cpp:198-235; caller guarantees padding here, no hidden
+  // allocations on the comm stream).
+  int64_t all_reduce_rsag(at::Tensor t, at::Tensor shard) {
+    check_flat(t);
+    check_flat(shard);
+    TORCH_CHECK(t.numel() == shard.numel() * size_, "bad rsag shapes");
+    RCCL_CHECK(ncclReduceScatter(t.data_ptr(), shard.data_ptr(), shard.numel(),
+                                 rccl_dtype(t), ncclSum, comm_,
+                                 stream_.stream()));
+    RCCL_CHECK(ncclAllGather(shard.data_ptr(), t.data_ptr(), shard.numel(),
+                             rccl_dtype(t), comm_, stream_.stream()));
+    return record();
+  }
+
+  int64_t send_recv(at::Tensor send, at::Tensor recv, int peer) {
+    check_flat(send);
+    check_flat(recv);
+    RCCL_CHECK(ncclGroupStart());
+    RCCL_CHECK(ncclSend(send.data_ptr(), send.numel(), rccl_dtype(send), peer,
+                        comm_, stream_.stream()));
+    RCCL_CHECK(ncclRecv(recv.data_ptr(), recv.numel(), rccl_dtype(recv), peer,
+                        comm_, stream_.stream()));
+    RCCL_CHECK(ncclGroupEnd());
+    return record();
+  }
+
+  // grouped broadcast of many tensors (parameter/state broadcast at startup)
+  int64_t broadcast_many(std::vector<at::Tensor> ts, int root) {
+    RCCL_CHECK(ncclGroupStart());
+    for (auto& t : ts) {
+      check_flat(t);
+      RCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
+                               rccl_dtype(t), root, comm_, stream_.stream()));
+    }
+    RCCL_CHECK(ncclGroupEnd());
+    return record();
+  }
+
+ private:
+  hipEvent_t acquire_event() {
+    std::lock_guard<std::mutex> g(mu_);
+    if (!event_pool_.empty()) {
+      hipEvent_t e = event_pool_.back();
+      event_pool_.pop_back();
+      return e;
+    }
+    hipEvent_t e;
+    HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+    return e;
+  }
+
+  void release_event(hipEvent_t e) {
+    std::lock_guard<std::mutex> g(mu_);
+    event_pool_.push_back(e);
+  }
+
+  int64_t record() {
+    hipEvent_t ev = acquire_event();
+    HIP_CHECK(hipEventRecord(ev, stream_.stream()));
+    std::lock_guard<std::mutex> g(mu_);
+    int64_t id = next_id_++;
+    live_events_[id] = ev;
+    // garbage-collect completed old events back into the pool
+    while (live_events_.size() > 64) {
+      auto it = live_events_.begin();
+      if (hipEventQuery(it->second) != hipSuccess) break;
+      event_pool_.push_back(it->second);
+      live_events_.erase(it);
+    }
+    return id;
+  }
+
+  int rank_, size_;
+  int device_;
+  c10::hip::HIPStream stream_;
+  ncclComm_t comm_ = nullptr;
+  std::mutex mu_;
+  int64_t next_id_ = 1;
+  std::map<int64_t, hipEvent_t> live_events_;
+  std::vector<hipEvent_t> event_pool_;
+};
+
+static py::bytes get_unique_id() {
+  ncclUniqueId uid;
+  RCCL_CHECK(ncclGetUniqueId(&uid));
+  return py::bytes(reinterpret_cast<const char*>(&uid), sizeof(uid));
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "DeAR MI355X-native RCCL communicator (hipEvent-handle design)";
+  m.def("get_unique_id", &get_unique_id);
+  py::class_<Communicator>(m, "Communicator")
+      .def(py::init<int, int, const std::string&>(), py::arg("rank"),
+           py::arg("size"), py::arg("uid"))
+      .def("rank", &Communicator::rank)
+      .def("size", &Communicator::size)
+      .def("stream_handle", &Communicator::stream_handle)
+      .def("wait_stream", &Communicator::wait_stream)
+      .def("wait_op_stream", &Communicator::wait_op_stream)
+      .def("wait_op_host", &Communicator::wait_op_host)
+      .def("op_done", &Communicator::op_done)
+      .def("synchronize", &Communicator::synchronize,
+           py::call_guard<py::gil_scoped_release>())
+      .def("all_reduce", &Communicator::all_reduce)
+      .def("reduce", &Communicator::reduce)
+      .def("broadcast", &Communicator::broadcast)
+      .def("reduce_scatter", &Communicator::reduce_scatter)
+      .def("all_gather", &Communicator::all_gather)
+      .def("all_reduce_rb", &Communicator::all_reduce_rb)
+      .def("all_reduce_rsag", &Communicator::all_reduce_rsag)
+      .def("send_recv", &Communicator::send_recv)
+      .def("broadcast_many", &Communicator::broadcast_many)
+      .def("wait_op_comm_static", &Communicator::wait_op_on);
+  m.def("wait_op_across",
+        [](Communicator& self, Communicator& other, int64_t id) {
+          Communicator::wait_op_on(self, other, id);
+        });
+}

@@ -1,0 +1,52 @@
+"""VGG-11/13/16/19 (the comm-bound stress model in the reference matrix:
+vgg16's 528 MB of gradients make DP sync the bottleneck)."""
+import torch.nn as nn
+
+__all__ = ["vgg11", "vgg13", "vgg16", "vgg19"]
+
+_CFG = {
+    "vgg11": [64, "M", 128, "M", 256, 256, "M", 512, 512, "M", 512, 512, "M"],
+    "vgg13": [64, 64, "M", 128, 128, "M", 256, 256, "M", 512, 512, "M",
+              512, 512, "M"],
+    "vgg16": [64, 64, "M", 128, 128, "M", 256, 256, 256, "M", 512, 512, 512,
+              "M", 512, 512, 512, "M"],
+    "vgg19": [64, 64, "M", 128, 128, "M", 256, 256, 256, 256, "M", 512, 512,
+              512, 512, "M", 512, 512, 512, 512, "M"],
+}
+
+
+class VGG(nn.Module):
+    def __init__(self, cfg, num_classes=1000, batch_norm=False):
+        super().__init__()
+        layers = []
+        cin = 3
+        for v in cfg:
+            if v == "M":
+                layers.append(nn.MaxPool2d(2, 2))
+            else:
+                layers.append(nn.Conv2d(cin, v, 3, padding=1))
+                if batch_norm:
+                    layers.append(nn.BatchNorm2d(v))
+                layers.append(nn.ReLU(inplace=True))
+                cin = v
+        self.features = nn.Sequential(*layers)
+        self.avgpool = nn.AdaptiveAvgPool2d(7)
+        self.classifier = nn.Sequential(
+            nn.Linear(512 * 7 * 7, 4096), nn.ReLU(True), nn.Dropout(),
+            nn.Linear(4096, 4096), nn.ReLU(True), nn.Dropout(),
+            nn.Linear(4096, num_classes))
+
+    def forward(self, x):
+        x = self.avgpool(self.features(x)).flatten(1)
+        return self.classifier(x)
+
+
+def _make(name):
+    def f(num_classes=1000, batch_norm=False):
+        return VGG(_CFG[name], num_classes, batch_norm)
+    f.__name__ = name
+    return f
+
+
+vgg11, vgg13, vgg16, vgg19 = (_make(n) for n in ("vgg11", "vgg13", "vgg16",
+                                                 "vgg19"))

@@ -142,6 +142,16 @@ class DearOptimizer(torch.optim.Optimizer):
         else:
             self._rs_handle[group.index] = NULL_HANDLE
 
+    def _enqueue_gather(self, g: BucketGroup):
+        """Enqueue the second half of the decoupled all-reduce for group g,
+        ordered device-side behind its reduce-scatter. Overridden by the
+        reduce+broadcast ablation."""
+        if self.size > 1 and self._do_ag and self._do_rs:
+            return self.comm_ag.all_gather(g.shard, g.bucket,
+                                           after=self._rs_handle[g.index])
+        # ablation / single-GPU: order behind RS only
+        return self._rs_handle[g.index]
+
     def _make_fw_hook(self, group: BucketGroup):
         def hook(module, inputs):
             if self._updated[group.index]:
@@ -179,13 +189,9 @@ class DearOptimizer(torch.optim.Optimizer):
             self._prev_iter_done.host_wait()
             last = NULL_HANDLE
             for g in self.groups:
-                if self.size > 1 and self._do_ag and self._do_rs:
-                    self._ag_handle[g.index] = self.comm_ag.all_gather(
-                        g.shard, g.bucket, after=self._rs_handle[g.index])
+                self._ag_handle[g.index] = self._enqueue_gather(g)
+                if self._ag_handle[g.index] is not NULL_HANDLE:
                     last = self._ag_handle[g.index]
-                else:
-                    # ablation / single-GPU: order behind RS only
-                    self._ag_handle[g.index] = self._rs_handle[g.index]
                 self._updated[g.index] = False
             self._prev_iter_done = last
             self._ready_count = [0] * len(self.groups)

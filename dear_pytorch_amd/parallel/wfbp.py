@@ -1,0 +1,112 @@
+"""WFBP / MG-WFBP baseline: wait-free backpropagation on the RCCL path.
+
+Reference capability: wfbp/dopt.py + mgwfbp/dopt.py — all-reduce each layer
+(or merged group) asynchronously from backward hooks as soon as its gradients
+are ready; synchronize and apply the wrapped optimizer in step().  Unlike
+DeAR there is no decoupling: step() blocks on every all-reduce.
+
+MI355X redesign: grad-as-bucket-view fused buffers (zero pack copies),
+all-reduce in place on the bucket via the side-stream RCCL communicator,
+average folded into the wrapped optimizer step (grads pre-divided by P with
+one fused scale pass).  MG-WFBP mode uses the measured alpha-beta merge
+planner (utils/perf_model.py) instead of a byte threshold.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from ..comm.backend import CommBackend, create_backend, NULL_HANDLE
+from .fusion import BucketGroup, build_groups
+
+__all__ = ["WfbpOptimizer"]
+
+
+class WfbpOptimizer(torch.optim.Optimizer):
+    def __init__(self, optimizer: torch.optim.Optimizer, model: torch.nn.Module,
+                 threshold_bytes: Optional[int] = None, mgwfbp: bool = False,
+                 fusion_flags: Optional[list] = None,
+                 backend: Optional[CommBackend] = None):
+        self.optim = optimizer
+        self.model = model
+        self.backend = backend or create_backend("wfbp")
+        self.rank, self.size = self.backend.rank, self.backend.size
+        self._device = next(model.parameters()).device
+        self._mgwfbp = mgwfbp
+        if mgwfbp and fusion_flags is None:
+            fusion_flags = self._plan_mgwfbp()
+        self.groups: List[BucketGroup] = build_groups(
+            model, threshold_bytes, fusion_flags=fusion_flags)
+        for g in self.groups:
+            g.allocate(self.size, self._device)
+        self._slot_of = {}
+        for g in self.groups:
+            for s in g.slots:
+                self._slot_of[s.param] = (g, s)
+        self._ready = [0] * len(self.groups)
+        self._handles = [NULL_HANDLE] * len(self.groups)
+        self._grad_accs = []
+        for g in self.groups:
+            for s in g.slots:
+                p = s.param
+                acc = p.expand_as(p).grad_fn.next_functions[0][0]
+                acc.register_hook(self._make_hook(p))
+                self._grad_accs.append(acc)
+
+    def _plan_mgwfbp(self):
+        """Measure per-layer backward times + fit xGMI alpha-beta, then merge
+        layers where waiting is cheaper than the saved startup alpha
+        (reference _generate_groups_mgwfbp, wfbp/dopt.py:380-486, with
+        MEASURED constants instead of the Ethernet tables)."""
+        from ..utils.perf_model import plan_mgwfbp_flags
+        return plan_mgwfbp_flags(self.model, self.backend)
+
+    def _make_hook(self, p):
+        def hook(*_):
+            group, slot = self._slot_of[p]
+            b = group.bucket[slot.offset: slot.offset + slot.numel]
+            if p.grad is not None and p.grad.data_ptr() != b.data_ptr():
+                b.view(p.shape).add_(p.grad)
+                p.grad = b.view(p.shape)
+            self._ready[group.index] += 1
+            if self._ready[group.index] == len(group.slots):
+                if self.size > 1:
+                    self._handles[group.index] = \
+                        self.backend.all_reduce(group.bucket)
+        return hook
+
+    def zero_grad(self, set_to_none: bool = False):
+        pass  # buckets are zeroed after each step below
+
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        for g in self.groups:
+            self._handles[g.index].wait_compute()
+            self._handles[g.index] = NULL_HANDLE
+        if self.size > 1:
+            for g in self.groups:
+                g.bucket.mul_(1.0 / self.size)
+        self.optim.step()
+        for g in self.groups:
+            g.bucket.zero_()
+        self._ready = [0] * len(self.groups)
+        return loss
+
+    def synchronize(self):
+        if torch.cuda.is_available() and self._device.type == "cuda":
+            torch.cuda.synchronize()
+
+    @property
+    def param_groups(self):
+        return self.optim.param_groups
+
+    @property
+    def state(self):
+        return self.optim.state
+
+    def state_dict(self):
+        return self.optim.state_dict()
+
+    def load_state_dict(self, sd):
+        self.optim.load_state_dict(sd)
