@@ -14,7 +14,13 @@ import json
 import os
 import time
 
-import torch
+# MIOpen's exhaustive conv search costs minutes per fresh box; FAST find uses
+# heuristics immediately (override with DEAR_MIOPEN_FIND=... if tuning runs
+# are wanted).
+os.environ.setdefault("MIOPEN_FIND_MODE", os.environ.get("DEAR_MIOPEN_FIND",
+                                                         "FAST"))
+
+import torch  # noqa: E402
 
 
 def parse_args():

@@ -19,6 +19,7 @@
 //   * errors surface as C++ exceptions (→ Python), not exit(1).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <c10/hip/HIPStream.h>
 #include <c10/hip/HIPGuard.h>
 #include <hip/hip_runtime.h>
@@ -73,10 +74,14 @@ void check_flat(const at::Tensor& t) {
 
 class Communicator {
  public:
+  // NOTE: the ROCm torch build masquerades HIP streams as CUDA-device streams;
+  // the pool accessor must be the MasqueradingAsCUDA variant or the
+  // HIPStream wrapper rejects the device type.
   Communicator(int rank, int size, const std::string& uid_bytes)
       : rank_(rank), size_(size),
         device_(c10::hip::current_device()),
-        stream_(c10::hip::getStreamFromPool(/*isHighPriority=*/true, device_)) {
+        stream_(at::hip::getStreamFromPoolMasqueradingAsCUDA(
+            /*isHighPriority=*/true, device_)) {
     TORCH_CHECK(uid_bytes.size() == sizeof(ncclUniqueId),
                 "bad rccl unique id size");
     ncclUniqueId uid;
@@ -270,7 +275,7 @@ class Communicator {
 
   int rank_, size_;
   int device_;
-  c10::hip::HIPStream stream_;
+  at::hip::HIPStreamMasqueradingAsCUDA stream_;
   ncclComm_t comm_ = nullptr;
   std::mutex mu_;
   int64_t next_id_ = 1;

@@ -1,0 +1,95 @@
+"""BO tuner, perf models, profilers (CPU)."""
+import os
+
+import torch
+import torch.nn as nn
+
+import dear_pytorch_amd as dear
+from dear_pytorch_amd.tuner import BayesOpt, ThresholdTuner
+from dear_pytorch_amd.profiling import (Profiling, CommunicationProfiler,
+                                        ChromeTracer)
+from dear_pytorch_amd.utils.perf_model import (AlphaBeta, fit_alpha_beta,
+                                               plan_mgwfbp_flags)
+
+
+def _model():
+    torch.manual_seed(0)
+    return nn.Sequential(nn.Linear(32, 64), nn.ReLU(), nn.Linear(64, 64),
+                         nn.ReLU(), nn.Linear(64, 8))
+
+
+def test_bayesopt_finds_peak():
+    bo = BayesOpt((0.0, 10.0))
+    f = lambda x: -(x - 7.0) ** 2  # max at 7
+    for _ in range(15):
+        x = bo.suggest()
+        bo.register(x, f(x))
+    best_x, _ = bo.best()
+    assert abs(best_x - 7.0) < 1.5
+
+
+def test_threshold_tuner_locks_and_regroups():
+    m = _model()
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.01), model=m,
+        threshold_bytes=1 << 14)
+    tuner = ThresholdTuner(opt, bounds_mb=(0.001, 0.1), window=2, warmup=2,
+                           trials=3, verbose=False)
+    x = torch.randn(4, 32)
+    y = torch.randn(4, 8)
+    for _ in range(20):
+        tuner.step_begin()
+        opt.zero_grad()
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+        tuner.step_end()
+    assert tuner.locked
+    # tuning must not break training numerics: params finite
+    assert all(torch.isfinite(p).all() for p in m.parameters())
+
+
+def test_fit_alpha_beta():
+    ab0 = AlphaBeta(alpha=5e-6, beta=2e-11)
+    sizes = [1 << s for s in range(14, 24)]
+    times = [ab0.alpha + ab0.beta * s for s in sizes]
+    fit = fit_alpha_beta(sizes, times)
+    assert abs(fit.alpha - ab0.alpha) / ab0.alpha < 0.2
+    assert abs(fit.beta - ab0.beta) / ab0.beta < 0.05
+
+
+def test_plan_mgwfbp_flags_shape():
+    m = _model()
+    flags = plan_mgwfbp_flags(m)
+    from dear_pytorch_amd.parallel.fusion import _module_param_order
+    assert len(flags) == len(_module_param_order(m))
+    assert flags[0] is True or flags[0] == 1.0
+
+
+def test_layerwise_profiling():
+    m = _model()
+    order, times, sizes = Profiling.benchmark(
+        m, lambda: (torch.randn(8, 32), torch.randn(8, 8)),
+        nn.functional.mse_loss, warmup=2, iters=4)
+    assert len(order) >= 1
+    assert all(t >= 0 for t in times.values())
+    assert set(sizes.values()) == {(32 * 64 + 64) * 4, (64 * 64 + 64) * 4,
+                                   (64 * 8 + 8) * 4}
+
+
+def test_comm_profiler_local():
+    from dear_pytorch_amd.comm.backend import LocalBackend
+    prof = CommunicationProfiler(LocalBackend(), sizes_bytes=[1 << 12, 1 << 14],
+                                 iters=2)
+    ab = prof.fit()
+    assert ab.alpha > 0 and ab.beta > 0
+
+
+def test_chrome_tracer(tmp_path):
+    p = tmp_path / "trace.json"
+    with ChromeTracer(str(p)) as t:
+        t.begin("fwd")
+        t.end("fwd")
+        t.instant("mark")
+    import json
+    ev = json.load(open(p))["traceEvents"]
+    assert [e["ph"] for e in ev] == ["B", "E", "i"]
