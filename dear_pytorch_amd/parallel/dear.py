@@ -82,9 +82,10 @@ class DearOptimizer(torch.optim.Optimizer):
         self._register_hooks()
 
     # ------------------------------------------------------------------ setup
-    def _build(self, threshold_bytes):
+    def _build(self, threshold_bytes, fusion_flags=None):
         self.groups: List[BucketGroup] = build_groups(
-            self.model, threshold_bytes, self.num_groups)
+            self.model, threshold_bytes, self.num_groups,
+            fusion_flags=fusion_flags)
         for g in self.groups:
             g.allocate(self.size, self._device)
         self._param_group_of = {}
@@ -211,7 +212,7 @@ class DearOptimizer(torch.optim.Optimizer):
     # regrouping (BO tuner / wait-time adaptive): rebuild buckets with a new
     # threshold between step() and the next forward (reference
     # dopt_rsag_bo.py:148-171 window).
-    def regroup(self, threshold_bytes: int):
+    def regroup(self, threshold_bytes: int = None, fusion_flags=None):
         self.synchronize()
         for h in self._hook_handles:
             h.remove()
@@ -219,8 +220,9 @@ class DearOptimizer(torch.optim.Optimizer):
         for g in self.groups:
             fused_ops.detach_group_state(self.optim, g)
             g.free()
-        self.threshold_bytes = threshold_bytes
-        self._build(threshold_bytes)
+        if threshold_bytes is not None:
+            self.threshold_bytes = threshold_bytes
+        self._build(self.threshold_bytes, fusion_flags=fusion_flags)
         self._register_hooks()
 
     # delegate the torch.optim.Optimizer surface to the wrapped optimizer

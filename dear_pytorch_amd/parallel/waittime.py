@@ -1,0 +1,106 @@
+"""Wait-time adaptive fusion (reference capability: dear/dopt_rsag_wt.py).
+
+Instead of a byte threshold, derive fusion boundaries from MEASURED
+wait-in-buffer times: start with everything merged, record how long each
+module's gradients sit in the bucket before the group completes during
+backward, then split groups so no gradient waits longer than a cycle budget
+(reference CYCLE_TIME = 5 ms, dopt_rsag_wt.py:40; EMA alpha = 0.9; flags
+broadcast from rank 0 at a fixed warmup step for cross-rank consistency).
+"""
+from __future__ import annotations
+
+import time
+from typing import Dict, List
+
+import torch
+
+__all__ = ["WaitTimeAdaptiveFusion"]
+
+
+class WaitTimeAdaptiveFusion:
+    def __init__(self, opt, cycle_time_s: float = 5e-3, ema: float = 0.9,
+                 regroup_at_step: int = 5, verbose: bool = True):
+        self.opt = opt
+        self.cycle = cycle_time_s
+        self.ema = ema
+        self.regroup_at = regroup_at_step
+        self.verbose = verbose and opt.rank == 0
+        self._step = 0
+        self._push_t: Dict[int, float] = {}
+        self._wait: Dict[int, float] = {}      # module-id -> EMA wait seconds
+        self._orig_hook = opt._make_bw_hook
+        self._done = False
+        self._install()
+
+    def _install(self):
+        """Wrap the optimizer's backward hooks to timestamp pushes and group
+        completions (capability of _update_wait_times, dopt_rsag_wt.py:376)."""
+        opt = self.opt
+        outer = self
+
+        def make_hook(p):
+            inner = outer._orig_hook(p)
+
+            def hook(*a):
+                outer._push_t[id(p)] = time.perf_counter()
+                inner(*a)
+                g, s = opt._slot_of[p]
+                if opt._ready_count[g.index] == len(g.slots):
+                    done = time.perf_counter()
+                    for slot in g.slots:
+                        t0 = outer._push_t.get(id(slot.param))
+                        if t0 is None:
+                            continue
+                        w = done - t0
+                        mid = id(slot.param)
+                        prev = outer._wait.get(mid, w)
+                        outer._wait[mid] = outer.ema * prev + \
+                            (1 - outer.ema) * w
+            return hook
+
+        opt._make_bw_hook = make_hook
+        # re-register so wrapped hooks are live (groups unchanged)
+        for h in opt._hook_handles:
+            h.remove()
+        opt._hook_handles.clear()
+        opt._register_hooks()
+
+    def step_end(self):
+        """Call once per training iteration, after opt.step()."""
+        self._step += 1
+        if self._done or self._step < self.regroup_at:
+            return
+        flags = self._flags_from_waits()
+        flags = self._sync_flags(flags)
+        self.opt._make_bw_hook = self._orig_hook  # stop timestamping
+        self.opt.regroup(fusion_flags=flags)
+        self._done = True
+        if self.verbose:
+            print(f"[dear-wt] regrouped into {len(self.opt.groups)} groups "
+                  f"at step {self._step}", flush=True)
+
+    def _flags_from_waits(self) -> List[bool]:
+        """Split where accumulated wait exceeds the cycle budget
+        (capability of _update_groups_with_wait_times,
+        dopt_rsag_wt.py:152-192)."""
+        from .fusion import _module_param_order
+        mods = _module_param_order(self.opt.model)
+        flags = [False] * len(mods)
+        if flags:
+            flags[0] = True
+        acc = 0.0
+        for i, (m, ps) in enumerate(mods):
+            w = max((self._wait.get(id(p), 0.0) for _, p in ps), default=0.0)
+            acc += w
+            if i > 0 and acc > self.cycle:
+                flags[i] = True
+                acc = 0.0
+        return flags
+
+    def _sync_flags(self, flags: List[bool]) -> List[bool]:
+        import torch.distributed as dist
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            t = torch.tensor([1.0 if f else 0.0 for f in flags])
+            dist.broadcast(t, src=0)
+            return [bool(v) for v in t.tolist()]
+        return flags
