@@ -39,6 +39,10 @@ def parse_args():
     p.add_argument("--threshold-mb", type=float, default=25.0)
     p.add_argument("--exclude-parts", default="")
     p.add_argument("--no-fusion", action="store_true")
+    p.add_argument("--channels-last", action="store_true", default=None,
+                   help="NHWC layout for CNNs (MIOpen igemm fast path; default on)")
+    p.add_argument("--no-channels-last", dest="channels_last",
+                   action="store_false")
     return p.parse_args()
 
 
@@ -79,6 +83,10 @@ def build_workload(args, device):
         res = 299 if args.model == "inceptionv4" else 224
         data = torch.randn(bs, 3, res, res, generator=g).to(device)
         target = torch.randint(0, 1000, (bs,), generator=g).to(device)
+        if args.channels_last or (args.channels_last is None and
+                                  device.type == "cuda"):
+            model = model.to(memory_format=torch.channels_last)
+            data = data.to(memory_format=torch.channels_last)
         lossf = torch.nn.CrossEntropyLoss().to(device)
 
         def step_fn(model, opt):

@@ -39,6 +39,15 @@ def _native():
     return _ext
 
 
+
+
+def _flat_param(p: torch.Tensor) -> torch.Tensor:
+    """Storage-order flat view of a dense (possibly channels_last) param."""
+    if p.is_contiguous():
+        return p.data.view(-1)
+    return p.data.as_strided((p.numel(),), (1,))
+
+
 def _kind(optim: torch.optim.Optimizer) -> str:
     import torch.optim as O
     t = type(optim)
@@ -193,7 +202,7 @@ def _sgd_python(group, hyp, scale, first_step):
     maximize = hyp.get("maximize", False)
     for s in group.slots:
         g = group.bucket[s.offset: s.offset + s.numel]
-        p = s.param.data.view(-1)
+        p = _flat_param(s.param)
         d_p = g.mul(scale)
         if maximize:
             d_p = -d_p
@@ -222,7 +231,7 @@ def _adam_python(group, hyp, scale, decoupled_wd, step):
     bc2 = 1.0 - b2 ** step
     for s in group.slots:
         g = group.bucket[s.offset: s.offset + s.numel]
-        p = s.param.data.view(-1)
+        p = _flat_param(s.param)
         grad = g.mul(scale)
         if decoupled_wd:
             p.mul_(1.0 - lr * wd)

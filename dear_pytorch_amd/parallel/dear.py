@@ -128,8 +128,10 @@ class DearOptimizer(torch.optim.Optimizer):
             # paths, replace .grad with a fresh tensor — detect and fold back.
             bucket_slice = group.bucket[slot.offset: slot.offset + slot.numel]
             if p.grad is not None and p.grad.data_ptr() != bucket_slice.data_ptr():
-                bucket_slice.view(p.shape).add_(p.grad)
-                p.grad = bucket_slice.view(p.shape)
+                from .fusion import grad_view
+                gv = grad_view(bucket_slice, p)
+                gv.add_(p.grad)
+                p.grad = gv
                 self._grad_view_fixups += 1
             self._ready_count[group.index] += 1
             if self._ready_count[group.index] == len(group.slots):

@@ -36,6 +36,31 @@ def _align(n: int, a: int = ALIGN_ELEMS) -> int:
     return (n + a - 1) // a * a
 
 
+def _is_dense(p: torch.Tensor) -> bool:
+    """True when p's strides are a permutation of a contiguous layout
+    (e.g. channels_last) — storage covers exactly numel elements."""
+    if p.is_contiguous():
+        return True
+    seen = sorted((s, d) for s, d in zip(p.stride(), p.shape) if d > 1)
+    expect = 1
+    for s, d in seen:
+        if s != expect:
+            return False
+        expect *= d
+    return True
+
+
+def grad_view(bucket_slice: torch.Tensor, p: torch.Tensor) -> torch.Tensor:
+    """View of the flat bucket slice with p's exact layout, so autograd
+    accumulates IN-PLACE into the bucket for contiguous AND channels_last
+    params (the fused kernels pair bucket[i] with p's storage element i,
+    which this layout match guarantees)."""
+    if p.is_contiguous():
+        return bucket_slice.view(p.shape)
+    assert _is_dense(p), "non-dense parameter layout unsupported"
+    return bucket_slice.as_strided(p.shape, p.stride())
+
+
 @dataclass
 class ParamSlot:
     name: str
@@ -66,13 +91,14 @@ class BucketGroup:
         shard_n = self.padded // world_size
         self.shard = torch.empty(shard_n, device=device, dtype=dtype)
         for s in self.slots:
-            g = self.bucket[s.offset: s.offset + s.numel].view(s.param.shape)
-            s.param.grad = g
+            s.param.grad = grad_view(
+                self.bucket[s.offset: s.offset + s.numel], s.param)
 
     def reattach_grads(self):
         """Re-point param.grad at bucket views (after anything detached them)."""
         for s in self.slots:
-            s.param.grad = self.bucket[s.offset: s.offset + s.numel].view(s.param.shape)
+            s.param.grad = grad_view(
+                self.bucket[s.offset: s.offset + s.numel], s.param)
 
     def free(self):
         for s in self.slots:

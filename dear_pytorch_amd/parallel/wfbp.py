@@ -67,8 +67,10 @@ class WfbpOptimizer(torch.optim.Optimizer):
             group, slot = self._slot_of[p]
             b = group.bucket[slot.offset: slot.offset + slot.numel]
             if p.grad is not None and p.grad.data_ptr() != b.data_ptr():
-                b.view(p.shape).add_(p.grad)
-                p.grad = b.view(p.shape)
+                from .fusion import grad_view
+                gv = grad_view(b, p)
+                gv.add_(p.grad)
+                p.grad = gv
             self._ready[group.index] += 1
             if self._ready[group.index] == len(group.slots):
                 if self.size > 1:

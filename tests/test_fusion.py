@@ -80,3 +80,33 @@ def test_allocate_grad_views_and_padding():
     assert torch.equal(
         g0.bucket[s0.offset: s0.offset + s0.numel].view(s0.param.shape),
         s0.param.grad)
+
+
+def test_channels_last_grad_views_accumulate_in_place():
+    import dear_pytorch_amd as dear
+    torch.manual_seed(0)
+    m = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(),
+                      nn.Conv2d(8, 4, 3, padding=1))
+    m = m.to(memory_format=torch.channels_last)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m)
+    x = torch.randn(2, 3, 8, 8).to(memory_format=torch.channels_last)
+    for _ in range(3):
+        opt.zero_grad()
+        m(x).sum().backward()
+        opt.step()
+    opt.synchronize()
+    assert opt._grad_view_fixups == 0, \
+        f"{opt._grad_view_fixups} out-of-place grad accumulations"
+    # numerics vs serial torch on the same layout
+    torch.manual_seed(0)
+    m2 = nn.Sequential(nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(),
+                       nn.Conv2d(8, 4, 3, padding=1))
+    m2 = m2.to(memory_format=torch.channels_last)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.05, momentum=0.9)
+    for _ in range(3):
+        o2.zero_grad()
+        m2(x).sum().backward()
+        o2.step()
+    for (na, pa), (_, pb) in zip(m2.named_parameters(), m.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), na
