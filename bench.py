@@ -39,6 +39,9 @@ def parse_args():
     p.add_argument("--threshold-mb", type=float, default=25.0)
     p.add_argument("--exclude-parts", default="")
     p.add_argument("--no-fusion", action="store_true")
+    p.add_argument("--comm-dtype", default="fp32",
+                   choices=["fp32", "bf16", "fp16"],
+                   help="gradient wire format (bf16/fp16 halve xGMI bytes)")
     p.add_argument("--channels-last", action="store_true", default=None,
                    help="NHWC layout for CNNs (MIOpen igemm fast path; default on)")
     p.add_argument("--no-channels-last", dest="channels_last",
@@ -111,9 +114,12 @@ def wrap_method(args, model, opt_fn):
         opt = opt_fn(model.parameters())
         return model, opt
     if args.method == "dear":
+        cdt = {"fp32": None, "bf16": torch.bfloat16,
+               "fp16": torch.float16}[args.comm_dtype]
         opt = dear.DistributedOptimizer(opt_fn(model.parameters()),
                                         model=model, threshold_bytes=threshold,
-                                        exclude_parts=args.exclude_parts)
+                                        exclude_parts=args.exclude_parts,
+                                        comm_dtype=cdt)
         return model, opt
     from dear_pytorch_amd.parallel import baselines
     opt = baselines.make(args.method, opt_fn(model.parameters()), model,

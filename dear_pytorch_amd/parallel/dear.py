@@ -66,6 +66,7 @@ class DearOptimizer(torch.optim.Optimizer):
         self._do_rs = "reducescatter" not in parts
         self._do_ag = "allgather" not in parts
 
+        self.comm_dtype = comm_dtype  # None => fp32 wire format
         if backend is None:
             backend = create_backend("dear_rs")
         self.comm_rs = backend
@@ -87,7 +88,8 @@ class DearOptimizer(torch.optim.Optimizer):
             self.model, threshold_bytes, self.num_groups,
             fusion_flags=fusion_flags)
         for g in self.groups:
-            g.allocate(self.size, self._device)
+            g.allocate(self.size, self._device,
+                       comm_dtype=self.comm_dtype if self.size > 1 else None)
         self._param_group_of = {}
         self._slot_of = {}
         for g in self.groups:
@@ -140,8 +142,13 @@ class DearOptimizer(torch.optim.Optimizer):
 
     def _launch_rs(self, group: BucketGroup):
         if self.size > 1 and self._do_rs:
+            wire = group.extra.get("comm_buf")
+            if wire is not None:
+                wire.copy_(group.bucket)  # fp32 -> bf16/fp16 cast, compute stream
+            else:
+                wire = group.bucket
             self._rs_handle[group.index] = self.comm_rs.reduce_scatter(
-                group.bucket, group.shard)
+                wire, group.shard)
         else:
             self._rs_handle[group.index] = NULL_HANDLE
 
@@ -150,7 +157,9 @@ class DearOptimizer(torch.optim.Optimizer):
         ordered device-side behind its reduce-scatter. Overridden by the
         reduce+broadcast ablation."""
         if self.size > 1 and self._do_ag and self._do_rs:
-            return self.comm_ag.all_gather(g.shard, g.bucket,
+            wire = g.extra.get("comm_buf")
+            return self.comm_ag.all_gather(g.shard,
+                                           wire if wire is not None else g.bucket,
                                            after=self._rs_handle[g.index])
         # ablation / single-GPU: order behind RS only
         return self._rs_handle[g.index]
@@ -170,6 +179,9 @@ class DearOptimizer(torch.optim.Optimizer):
     def _apply_update(self, group: BucketGroup):
         """Averaged-grad optimizer step for every param in `group`, then re-zero
         the bucket so backward can accumulate fresh gradients into it."""
+        wire = group.extra.get("comm_buf")
+        if wire is not None and self.size > 1 and self._do_rs and self._do_ag:
+            group.bucket.copy_(wire)  # bf16/fp16 -> fp32 cast back
         fused_ops.fused_group_step(self.optim, group, self.size,
                                    apply_ag=self._do_ag)
 

@@ -137,3 +137,35 @@ def test_broadcast_optimizer_state_ws2():
     assert a.keys() == b.keys() and len(a) > 0
     for k in a:
         assert abs(a[k] - b[k]) < 1e-6, k
+
+
+def _rank_train_bf16comm(rank, world, T, bs):
+    import torch
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m,
+        threshold_bytes=1 << 12, comm_dtype=torch.float16)
+    for x, y in _full_data(T, bs):
+        xs, ys = x[rank * bs:(rank + 1) * bs], y[rank * bs:(rank + 1) * bs]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    opt.synchronize()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_dear_ws2_fp16_comm_close_to_serial():
+    T, bs = 5, 8
+    ref = _serial_reference(T, bs)
+    outs = run_dist(_rank_train_bf16comm, world_size=2, args=(T, bs))
+    for k in ref:
+        # reduced-precision wire: looser tolerance, but must track
+        assert torch.allclose(ref[k], outs[0][k], atol=5e-3), \
+            f"{k}: {(ref[k] - outs[0][k]).abs().max():.3e}"
+        assert torch.equal(outs[0][k], outs[1][k])
