@@ -61,6 +61,9 @@ def main():
     data = torch.randn(args.batch_size, 3, res, res, generator=g).to(device)
     target = torch.randint(0, 1000, (args.batch_size,), generator=g).to(device)
     lossf = torch.nn.CrossEntropyLoss().to(device)
+    if on_gpu:  # NHWC: MIOpen's fast igemm path on CDNA (profiles/README.md)
+        model = model.to(memory_format=torch.channels_last)
+        data = data.to(memory_format=torch.channels_last)
 
     if world > 1:
         dear.broadcast_parameters(model.state_dict(), root_rank=0)
@@ -81,8 +84,12 @@ def main():
             tuner = ThresholdTuner(opt)
     else:
         from dear_pytorch_amd.parallel import baselines
+        kw = {}
+        if args.method in ("wfbp", "mgwfbp") and args.compressor != "none" \
+                and args.density < 1.0:
+            kw = dict(compressor=args.compressor, density=args.density)
         opt = baselines.make(args.method, base_opt, model,
-                             threshold_bytes=threshold)
+                             threshold_bytes=threshold, **kw)
 
     model.train()
 

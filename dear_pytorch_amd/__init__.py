@@ -86,10 +86,14 @@ def allreduce(tensor: torch.Tensor, average: bool = True, name: str | None = Non
     """Blocking mean/sum all-reduce of a (typically small metric) tensor —
     reference dear/dopt_rsag.py:543."""
     t = tensor.detach().clone()
+    src_device = t.device
+    if (size() > 1 and torch.cuda.is_available()
+            and dist.get_backend() == "nccl" and not t.is_cuda):
+        t = t.cuda()  # native RCCL channel needs device tensors
     _generic().all_reduce(t).host_wait()
     if average and size() > 1:
         t /= size()
-    return t
+    return t.to(src_device)
 
 
 def broadcast_parameters(params, root_rank: int = 0):

@@ -19,6 +19,9 @@ def make(method: str, optimizer, model, threshold_bytes=None, **kw):
     if method == "wfbp":
         # classic WFBP: per-layer (no fusion) all-reduce from backward hooks
         return WfbpOptimizer(optimizer, model, threshold_bytes=None, **kw)
+    if method == "wfbp-fused":
+        return WfbpOptimizer(optimizer, model, threshold_bytes=threshold_bytes,
+                             **kw)
     if method == "mgwfbp":
         return WfbpOptimizer(optimizer, model, threshold_bytes=threshold_bytes,
                              mgwfbp=True, **kw)

@@ -27,13 +27,22 @@ class WfbpOptimizer(torch.optim.Optimizer):
     def __init__(self, optimizer: torch.optim.Optimizer, model: torch.nn.Module,
                  threshold_bytes: Optional[int] = None, mgwfbp: bool = False,
                  fusion_flags: Optional[list] = None,
-                 backend: Optional[CommBackend] = None):
+                 backend: Optional[CommBackend] = None,
+                 compressor: Optional[str] = None, density: float = 1.0):
         self.optim = optimizer
         self.model = model
         self.backend = backend or create_backend("wfbp")
         self.rank, self.size = self.backend.rank, self.backend.size
         self._device = next(model.parameters()).device
         self._mgwfbp = mgwfbp
+        # sparsified sync (reference wfbp/dopt.py compression machinery):
+        # top-k style codecs communicate (values, indices) via all-gather
+        self.density = density
+        self.compressor = None
+        if compressor not in (None, "none") and density < 1.0:
+            from ..compression import compressors
+            self.compressor = compressors[compressor]()
+        self._sparse_ctx = {}
         if mgwfbp and fusion_flags is None:
             fusion_flags = self._plan_mgwfbp()
         self.groups: List[BucketGroup] = build_groups(
@@ -74,9 +83,26 @@ class WfbpOptimizer(torch.optim.Optimizer):
             self._ready[group.index] += 1
             if self._ready[group.index] == len(group.slots):
                 if self.size > 1:
-                    self._handles[group.index] = \
-                        self.backend.all_reduce(group.bucket)
+                    if self.compressor is not None:
+                        self._launch_sparse(group)
+                    else:
+                        self._handles[group.index] = \
+                            self.backend.all_reduce(group.bucket)
         return hook
+
+    def _launch_sparse(self, group):
+        """Sparsified sync: compress to (values, indices), all-gather both
+        (reference wfbp/dopt.py:732-738), reassemble in step()."""
+        flat = group.bucket
+        _, (vals, idx) = self.compressor.compress(
+            flat, name=f"g{group.index}", ratio=self.density)
+        k = vals.numel()
+        av = torch.empty(k * self.size, device=vals.device, dtype=vals.dtype)
+        ai = torch.empty(k * self.size, device=idx.device, dtype=idx.dtype)
+        h1 = self.backend.all_gather(vals.contiguous(), av)
+        h2 = self.backend.all_gather(idx.contiguous(), ai)
+        self._sparse_ctx[group.index] = (av, ai)
+        self._handles[group.index] = (h1, h2)
 
     def zero_grad(self, set_to_none: bool = False):
         pass  # buckets are zeroed after each step below
@@ -84,8 +110,17 @@ class WfbpOptimizer(torch.optim.Optimizer):
     def step(self, closure=None):
         loss = closure() if closure is not None else None
         for g in self.groups:
-            self._handles[g.index].wait_compute()
+            h = self._handles[g.index]
+            if isinstance(h, tuple):
+                for hh in h:
+                    hh.wait_compute()
+            else:
+                h.wait_compute()
             self._handles[g.index] = NULL_HANDLE
+            if g.index in self._sparse_ctx:
+                av, ai = self._sparse_ctx.pop(g.index)
+                g.bucket.zero_()
+                g.bucket.scatter_add_(0, ai, av)
         if self.size > 1:
             for g in self.groups:
                 g.bucket.mul_(1.0 / self.size)

@@ -169,3 +169,59 @@ def test_dear_ws2_fp16_comm_close_to_serial():
         assert torch.allclose(ref[k], outs[0][k], atol=5e-3), \
             f"{k}: {(ref[k] - outs[0][k]).abs().max():.3e}"
         assert torch.equal(outs[0][k], outs[1][k])
+
+
+def _rank_wfbp_sparse(rank, world):
+    import torch
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    from dear_pytorch_amd.parallel.wfbp import WfbpOptimizer
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = WfbpOptimizer(torch.optim.SGD(m.parameters(), lr=0.05), m,
+                        threshold_bytes=1 << 12, compressor="topk",
+                        density=0.25)
+    for x, y in _full_data(4, 8):
+        xs, ys = x[rank * 8:(rank + 1) * 8], y[rank * 8:(rank + 1) * 8]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_wfbp_sparse_topk_ws2_ranks_agree():
+    outs = run_dist(_rank_wfbp_sparse, world_size=2)
+    for k in outs[0]:
+        assert torch.equal(outs[0][k], outs[1][k]), k
+        assert torch.isfinite(outs[0][k]).all()
+
+
+def _rank_wfbp_dense(rank, world):
+    import torch
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    from dear_pytorch_amd.parallel.wfbp import WfbpOptimizer
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = WfbpOptimizer(torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9),
+                        m, threshold_bytes=None)
+    for x, y in _full_data(5, 8):
+        xs, ys = x[rank * 8:(rank + 1) * 8], y[rank * 8:(rank + 1) * 8]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_wfbp_dense_ws2_matches_serial():
+    ref = _serial_reference(5, 8)
+    outs = run_dist(_rank_wfbp_dense, world_size=2)
+    for k in ref:
+        assert torch.allclose(ref[k], outs[0][k], atol=1e-5), k
+        assert torch.equal(outs[0][k], outs[1][k])
