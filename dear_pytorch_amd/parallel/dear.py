@@ -103,9 +103,11 @@ class DearOptimizer(torch.optim.Optimizer):
         self._updated = [True] * n   # True => no pending gathered grads to apply
         self._prev_iter_done = NULL_HANDLE
         if self.rank == 0 and os.environ.get("DEAR_QUIET", "0") != "1":
+            import sys
             mb = [g.nbytes / 1e6 for g in self.groups]
             print(f"[dear] {n} fusion groups, sizes MB: "
-                  f"{', '.join(f'{m:.1f}' for m in mb)}", flush=True)
+                  f"{', '.join(f'{m:.1f}' for m in mb)}", file=sys.stderr,
+                  flush=True)
 
     def _register_hooks(self):
         # per-parameter grad-accumulator hooks -> reduce-scatter on group-complete
