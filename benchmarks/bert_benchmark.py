@@ -103,10 +103,18 @@ def main():
         f"{world} GPU(s)")
     timeit.timeit(benchmark_step, number=args.num_warmup_batches)
     sen_secs = []
+    import time as _time
     for _ in range(args.num_iters):
-        t = timeit.timeit(benchmark_step, number=args.num_batches_per_iter)
+        # sync-bracketed window: DeAR's step() is async, so honest timing
+        # needs device completion inside the window (unlike host timeit)
         if on_gpu:
             torch.cuda.synchronize()
+        t0 = _time.perf_counter()
+        for _b in range(args.num_batches_per_iter):
+            benchmark_step()
+        if on_gpu:
+            torch.cuda.synchronize()
+        t = _time.perf_counter() - t0
         sen_secs.append(bs * args.num_batches_per_iter / t)
     mean = np.mean(sen_secs)
     conf = 1.96 * np.std(sen_secs)

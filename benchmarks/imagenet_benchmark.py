@@ -107,10 +107,18 @@ def main():
         f"{world} GPU(s)")
     timeit.timeit(benchmark_step, number=args.num_warmup_batches)
     img_secs = []
+    import time as _time
     for _ in range(args.num_iters):
-        t = timeit.timeit(benchmark_step, number=args.num_batches_per_iter)
+        # sync-bracketed window: DeAR's step() is async, so honest timing
+        # needs device completion inside the window (unlike host timeit)
         if on_gpu:
             torch.cuda.synchronize()
+        t0 = _time.perf_counter()
+        for _b in range(args.num_batches_per_iter):
+            benchmark_step()
+        if on_gpu:
+            torch.cuda.synchronize()
+        t = _time.perf_counter() - t0
         img_secs.append(args.batch_size * args.num_batches_per_iter / t)
     img_sec_mean = np.mean(img_secs)
     img_sec_conf = 1.96 * np.std(img_secs)
