@@ -34,7 +34,7 @@ def parse_args():
                    help="per-GPU batch (default: 64 CNN / 32 BERT)")
     p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--method", default="dear",
-                   choices=["dear", "ddp", "wfbp", "mgwfbp", "naive", "rb"],
+                   choices=["dear", "ddp", "wfbp", "mgwfbp", "naive", "rb", "bytescheduler"],
                    help="gradient-sync method (dear is the product)")
     p.add_argument("--threshold-mb", type=float, default=25.0)
     p.add_argument("--exclude-parts", default="")

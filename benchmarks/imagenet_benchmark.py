@@ -30,7 +30,7 @@ def main():
     p.add_argument("--num-iters", type=int, default=5)
     p.add_argument("--method", default="dear",
                    choices=["dear", "dear-bo", "ddp", "wfbp", "mgwfbp",
-                            "naive", "rb"])
+                            "naive", "rb", "bytescheduler"])
     p.add_argument("--threshold", type=int, default=25 * 1024 * 1024,
                    help="fusion threshold bytes")
     p.add_argument("--no-fusion", action="store_true")

@@ -10,6 +10,7 @@ from __future__ import annotations
 from .wfbp import WfbpOptimizer
 from .naive import NaiveDearOptimizer
 from .rb import ReduceBcastOptimizer
+from .bytescheduler import ByteSchedulerOptimizer
 
 __all__ = ["make"]
 
@@ -27,6 +28,8 @@ def make(method: str, optimizer, model, threshold_bytes=None, **kw):
                              mgwfbp=True, **kw)
     if method == "naive":
         return NaiveDearOptimizer(optimizer, model, **kw)
+    if method == "bytescheduler":
+        return ByteSchedulerOptimizer(optimizer, model, **kw)
     if method == "rb":
         return ReduceBcastOptimizer(optimizer, model,
                                     threshold_bytes=threshold_bytes, **kw)

@@ -225,3 +225,32 @@ def test_wfbp_dense_ws2_matches_serial():
     for k in ref:
         assert torch.allclose(ref[k], outs[0][k], atol=1e-5), k
         assert torch.equal(outs[0][k], outs[1][k])
+
+
+def _rank_bytescheduler(rank, world):
+    import torch
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    from dear_pytorch_amd.parallel.bytescheduler import ByteSchedulerOptimizer
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = ByteSchedulerOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), m,
+        partition_bytes=256)
+    for x, y in _full_data(5, 8):
+        xs, ys = x[rank * 8:(rank + 1) * 8], y[rank * 8:(rank + 1) * 8]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_bytescheduler_ws2_matches_serial():
+    ref = _serial_reference(5, 8)
+    outs = run_dist(_rank_bytescheduler, world_size=2)
+    for k in ref:
+        assert torch.allclose(ref[k], outs[0][k], atol=1e-5), k
+        assert torch.equal(outs[0][k], outs[1][k])
