@@ -29,7 +29,7 @@ def main():
     p.add_argument("--num-batches-per-iter", type=int, default=10)
     p.add_argument("--num-iters", type=int, default=5)
     p.add_argument("--method", default="dear",
-                   choices=["dear", "dear-bo", "ddp", "wfbp", "mgwfbp",
+                   choices=["dear", "dear-bo", "dear-wt", "ddp", "wfbp", "mgwfbp",
                             "naive", "rb", "bytescheduler"])
     p.add_argument("--threshold", type=int, default=25 * 1024 * 1024,
                    help="fusion threshold bytes")
@@ -75,13 +75,25 @@ def main():
             model = torch.nn.parallel.DistributedDataParallel(
                 model, gradient_as_bucket_view=True)
         opt = base_opt
-    elif args.method in ("dear", "dear-bo"):
+    elif args.method in ("dear", "dear-bo", "dear-wt"):
         opt = dear.DistributedOptimizer(base_opt, model=model,
                                         threshold_bytes=threshold,
                                         exclude_parts=args.exclude_parts)
         if args.method == "dear-bo":
             from dear_pytorch_amd.tuner import ThresholdTuner
             tuner = ThresholdTuner(opt)
+        elif args.method == "dear-wt":
+            from dear_pytorch_amd.parallel.waittime import \
+                WaitTimeAdaptiveFusion
+            wt = WaitTimeAdaptiveFusion(opt)
+
+            class _WtTuner:  # same step_begin/step_end surface as the BO tuner
+                def step_begin(self):
+                    pass
+
+                def step_end(self):
+                    wt.step_end()
+            tuner = _WtTuner()
     else:
         from dear_pytorch_amd.parallel import baselines
         kw = {}

@@ -78,6 +78,8 @@ class DearOptimizer(torch.optim.Optimizer):
         self._num_steps = 0
         self._hook_handles = []
         self._grad_view_fixups = 0
+        from ..profiling import tracer
+        self._tracer = tracer()  # None unless DEAR_TIMELINE is set
 
         self._build(threshold_bytes)
         self._register_hooks()
@@ -139,6 +141,8 @@ class DearOptimizer(torch.optim.Optimizer):
                 self._grad_view_fixups += 1
             self._ready_count[group.index] += 1
             if self._ready_count[group.index] == len(group.slots):
+                if self._tracer:
+                    self._tracer.instant(f"rs_launch/g{group.index}", "comm")
                 self._launch_rs(group)
         return hook
 
@@ -172,8 +176,12 @@ class DearOptimizer(torch.optim.Optimizer):
                 return
             # wait (device-side) for this group's all-gather, then apply the
             # lazy fused update, pipelined with forward compute of later groups
+            if self._tracer:
+                self._tracer.begin(f"ag_wait+update/g{group.index}", "comm")
             self._ag_handle[group.index].wait_compute()
             self._apply_update(group)
+            if self._tracer:
+                self._tracer.end(f"ag_wait+update/g{group.index}", "comm")
             self._updated[group.index] = True
         return hook
 

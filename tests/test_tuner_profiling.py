@@ -118,3 +118,35 @@ def test_waittime_adaptive_fusion_regroups():
         opt.step()
     opt.synchronize()
     assert all(torch.isfinite(p).all() for p in m.parameters())
+
+
+def test_checkpoint_save_load_roundtrip(tmp_path):
+    from dear_pytorch_amd import checkpoint
+    m = _model()
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m)
+    x, y = torch.randn(4, 32), torch.randn(4, 8)
+    for _ in range(3):
+        opt.zero_grad()
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+    p = str(tmp_path / "ck.pt")
+    checkpoint.save(p, m, opt, step=3)
+    m2 = _model()
+    opt2 = dear.DistributedOptimizer(
+        torch.optim.SGD(m2.parameters(), lr=0.05, momentum=0.9), model=m2)
+    step = checkpoint.load(p, m2, opt2.optim)
+    assert step == 3
+    for (ka, va), (kb, vb) in zip(m.state_dict().items(),
+                                  m2.state_dict().items()):
+        assert torch.equal(va, vb), ka
+    # resumed training continues identically
+    for _ in range(2):
+        for opt_, mm in ((opt, m), (opt2, m2)):
+            opt_.zero_grad()
+            nn.functional.mse_loss(mm(x), y).backward()
+            opt_.step()
+    opt.synchronize(); opt2.synchronize()
+    for (ka, va), (_, vb) in zip(m.state_dict().items(),
+                                 m2.state_dict().items()):
+        assert torch.allclose(va, vb, atol=1e-6), ka
