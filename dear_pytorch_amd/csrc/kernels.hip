@@ -134,22 +134,52 @@ __global__ __launch_bounds__(kBlock) void fused_adam_kernel(
   float* __restrict__ g = bucket + off;
   float* __restrict__ m = ma + off;
   float* __restrict__ v = va + off;
-  for (int i = threadIdx.x; i < n; i += kBlock) {
-    float gj = g[i] * scale;
-    float pj = p[i];
+
+  auto upd = [&](float gj, float pj, float& mj, float& vj) -> float {
+    gj *= scale;
     if (kDecoupled) {
       pj *= (1.f - lr * wd);
     } else if (wd != 0.f) {
       gj = fmaf(wd, pj, gj);
     }
-    float mj = fmaf(b1, m[i], (1.f - b1) * gj);
-    float vj = fmaf(b2, v[i], (1.f - b2) * gj * gj);
-    m[i] = mj;
-    v[i] = vj;
+    mj = fmaf(b1, mj, (1.f - b1) * gj);
+    vj = fmaf(b2, vj, (1.f - b2) * gj * gj);
     // p -= lr/bc1 * m / (sqrt(v)/sqrt(bc2) + eps)   (torch.optim.Adam order)
     float denom = fmaf(sqrtf(vj), inv_sqrt_bc2, eps);
-    p[i] = fmaf(-(lr * inv_bc1), mj / denom, pj);
-    g[i] = 0.f;
+    return fmaf(-(lr * inv_bc1), mj / denom, pj);
+  };
+
+  if (aligned16(p) && aligned16(g) && aligned16(m) && aligned16(v)) {
+    const int n4 = n >> 2;
+    float4* p4 = reinterpret_cast<float4*>(p);
+    float4* g4 = reinterpret_cast<float4*>(g);
+    float4* m4 = reinterpret_cast<float4*>(m);
+    float4* v4 = reinterpret_cast<float4*>(v);
+    for (int i = threadIdx.x; i < n4; i += kBlock) {
+      float4 gv = g4[i], pv = p4[i], mv = m4[i], vv = v4[i];
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        (&pv.x)[j] = upd((&gv.x)[j], (&pv.x)[j], (&mv.x)[j], (&vv.x)[j]);
+      p4[i] = pv;
+      m4[i] = mv;
+      v4[i] = vv;
+      g4[i] = make_float4(0.f, 0.f, 0.f, 0.f);
+    }
+    for (int i = (n4 << 2) + threadIdx.x; i < n; i += kBlock) {
+      float mj = m[i], vj = v[i];
+      p[i] = upd(g[i], p[i], mj, vj);
+      m[i] = mj;
+      v[i] = vj;
+      g[i] = 0.f;
+    }
+  } else {
+    for (int i = threadIdx.x; i < n; i += kBlock) {
+      float mj = m[i], vj = v[i];
+      p[i] = upd(g[i], p[i], mj, vj);
+      m[i] = mj;
+      v[i] = vj;
+      g[i] = 0.f;
+    }
   }
 }
 
