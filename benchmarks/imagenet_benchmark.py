@@ -100,6 +100,14 @@ def main():
         if args.method in ("wfbp", "mgwfbp") and args.compressor != "none" \
                 and args.density < 1.0:
             kw = dict(compressor=args.compressor, density=args.density)
+        if args.method == "mgwfbp":
+            # reference protocol: measure per-layer backward times first and
+            # hand them to the planner (mgwfbp/imagenet_benchmark.py:98-100)
+            from dear_pytorch_amd.profiling import Profiling
+            _, times, _ = Profiling.benchmark(
+                model, lambda: (data, target),
+                lambda out, tgt: lossf(out, tgt), warmup=2, iters=5)
+            kw["layerwise_times"] = times
         opt = baselines.make(args.method, base_opt, model,
                              threshold_bytes=threshold, **kw)
 

@@ -28,13 +28,15 @@ class WfbpOptimizer(torch.optim.Optimizer):
                  threshold_bytes: Optional[int] = None, mgwfbp: bool = False,
                  fusion_flags: Optional[list] = None,
                  backend: Optional[CommBackend] = None,
-                 compressor: Optional[str] = None, density: float = 1.0):
+                 compressor: Optional[str] = None, density: float = 1.0,
+                 layerwise_times=None):
         self.optim = optimizer
         self.model = model
         self.backend = backend or create_backend("wfbp")
         self.rank, self.size = self.backend.rank, self.backend.size
         self._device = next(model.parameters()).device
         self._mgwfbp = mgwfbp
+        self._layerwise_times = layerwise_times
         # sparsified sync (reference wfbp/dopt.py compression machinery):
         # top-k style codecs communicate (values, indices) via all-gather
         self.density = density
@@ -69,7 +71,19 @@ class WfbpOptimizer(torch.optim.Optimizer):
         (reference _generate_groups_mgwfbp, wfbp/dopt.py:380-486, with
         MEASURED constants instead of the Ethernet tables)."""
         from ..utils.perf_model import plan_mgwfbp_flags
-        return plan_mgwfbp_flags(self.model, self.backend)
+        ab = None
+        if self.size > 1 and self._device.type == "cuda":
+            # fit alpha-beta on the live xGMI fabric (reference fits with
+            # CommunicationProfiler + LinearRegression, wfbp/dopt.py:260-285)
+            from ..profiling import CommunicationProfiler
+            ab = CommunicationProfiler(self.backend, iters=5).fit()
+            if self.rank == 0:
+                import sys
+                print(f"[mgwfbp] measured alpha={ab.alpha * 1e6:.1f}us "
+                      f"beta={1.0 / ab.beta / 1e9:.1f}GB/s", file=sys.stderr,
+                      flush=True)
+        return plan_mgwfbp_flags(self.model, self.backend,
+                                 layerwise_times=self._layerwise_times, ab=ab)
 
     def _make_hook(self, p):
         def hook(*_):
