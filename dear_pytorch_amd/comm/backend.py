@@ -242,6 +242,8 @@ class RcclBackend(CommBackend):
     rings use xGMI link capacity concurrently (7 p2p links/GPU).
     """
 
+    _instances = 0  # SPMD-ordered creation => identical count on every rank
+
     def __init__(self, device: torch.device, tag: str = "generic"):
         import dear_pytorch_amd._comm_core as comm_core  # loud ImportError on GPU box
 
@@ -250,8 +252,10 @@ class RcclBackend(CommBackend):
         self.size = dist.get_world_size()
         self.device = device
         # RCCL unique-id exchange over the torch.distributed store (replaces
-        # the reference's MPI_Bcast, communicator.cpp:43-57).
-        uid_key = f"dear_rccl_uid/{tag}"
+        # the reference's MPI_Bcast, communicator.cpp:43-57).  The instance
+        # counter keeps keys unique when several backends reuse a tag.
+        RcclBackend._instances += 1
+        uid_key = f"dear_rccl_uid/{tag}/{RcclBackend._instances}"
         store = dist.distributed_c10d._get_default_store()
         if self.rank == 0:
             uid = comm_core.get_unique_id()
