@@ -42,6 +42,10 @@ def parse_args():
     p.add_argument("--comm-dtype", default="fp32",
                    choices=["fp32", "bf16", "fp16"],
                    help="gradient wire format (bf16/fp16 halve xGMI bytes)")
+    p.add_argument("--fused-bn", action="store_true", default=None,
+                   help="fused NHWC BN+ReLU+residual kernels for ResNets "
+                        "(default on for GPU ResNets)")
+    p.add_argument("--no-fused-bn", dest="fused_bn", action="store_false")
     p.add_argument("--channels-last", action="store_true", default=None,
                    help="NHWC layout for CNNs (MIOpen igemm fast path; default on)")
     p.add_argument("--no-channels-last", dest="channels_last",
@@ -81,7 +85,10 @@ def build_workload(args, device):
         unit, metric = "sen/sec", "sen/sec"
     else:
         bs = args.batch_size or 64
-        model = models.get_cnn(args.model).to(device)
+        fused_bn = args.fused_bn
+        if fused_bn is None:
+            fused_bn = device.type == "cuda" and args.model.startswith("resnet")
+        model = models.get_cnn(args.model, fused_bn=fused_bn).to(device)
         g = torch.Generator().manual_seed(1234)
         res = 299 if args.model == "inceptionv4" else 224
         data = torch.randn(bs, 3, res, res, generator=g).to(device)

@@ -356,8 +356,56 @@ static void select_ge(at::Tensor x, double thr, at::Tensor out_idx,
   HIP_CHECK(hipGetLastError());
 }
 
+// ---------------------------------------------------------- fused BN (NHWC)
+extern "C" void dear_bn_fwd(hipStream_t, const float*, const float*, float*,
+                            const float*, const float*, float*, float*,
+                            float*, float*, float*, float*, int, long, int,
+                            float, float, int, int);
+extern "C" void dear_bn_bwd(hipStream_t, const float*, const float*,
+                            const float*, float*, const float*, const float*,
+                            const float*, float*, float*, int, float*, float*,
+                            float*, long, int, int, int);
+
+static void bn_fwd(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor y,
+                   at::Tensor w, at::Tensor b, at::Tensor mean,
+                   at::Tensor invstd, at::Tensor rmean, at::Tensor rvar,
+                   at::Tensor psum, at::Tensor psumsq, int64_t rows, int64_t C,
+                   double eps, double momentum, bool training, bool relu) {
+  auto stream = c10::hip::getCurrentHIPStream();
+  dear_bn_fwd(stream.stream(), x.data_ptr<float>(),
+              res ? res->data_ptr<float>() : nullptr, y.data_ptr<float>(),
+              w.data_ptr<float>(), b.data_ptr<float>(),
+              mean.data_ptr<float>(), invstd.data_ptr<float>(),
+              rmean.data_ptr<float>(), rvar.data_ptr<float>(),
+              psum.data_ptr<float>(), psumsq.data_ptr<float>(),
+              (int)psum.size(0), (long)rows, (int)C, (float)eps,
+              (float)momentum, training ? 1 : 0, relu ? 1 : 0);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
+                   c10::optional<at::Tensor> dy_eff, at::Tensor w,
+                   at::Tensor mean, at::Tensor invstd, at::Tensor pdb,
+                   at::Tensor pdg, at::Tensor dbeta, at::Tensor dgamma,
+                   at::Tensor dx, int64_t rows, int64_t C, bool relu,
+                   bool want_dy_eff) {
+  auto stream = c10::hip::getCurrentHIPStream();
+  dear_bn_bwd(stream.stream(), x.data_ptr<float>(), dy.data_ptr<float>(),
+              y.data_ptr<float>(),
+              dy_eff ? dy_eff->data_ptr<float>() : nullptr,
+              w.data_ptr<float>(), mean.data_ptr<float>(),
+              invstd.data_ptr<float>(), pdb.data_ptr<float>(),
+              pdg.data_ptr<float>(), (int)pdb.size(0),
+              dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
+              dx.data_ptr<float>(), (long)rows, (int)C, relu ? 1 : 0,
+              want_dy_eff ? 1 : 0);
+  HIP_CHECK(hipGetLastError());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "DeAR CDNA4 fused kernels (gfx950)";
+  m.def("bn_fwd", &bn_fwd);
+  m.def("bn_bwd", &bn_bwd);
   m.def("fused_sgd", &fused_sgd);
   m.def("fused_adam", &fused_adam);
   m.def("pack", &pack);

@@ -23,12 +23,14 @@ _REGISTRY = {
 }
 
 
-def get_cnn(name: str, num_classes: int = 1000):
+def get_cnn(name: str, num_classes: int = 1000, fused_bn: bool = False):
     name = name.lower()
     if name not in _REGISTRY:
         raise KeyError(f"unknown model '{name}'; have {sorted(_REGISTRY)}")
     if name == "mnistnet":
         return _REGISTRY[name]()
+    if name.startswith("resnet"):
+        return _REGISTRY[name](num_classes=num_classes, fused_bn=fused_bn)
     return _REGISTRY[name](num_classes=num_classes)
 
 

@@ -1,0 +1,106 @@
+"""Fused NHWC BatchNorm(+ReLU)(+residual) training op (CDNA4 kernels).
+
+``FusedBNAct2d`` subclasses nn.BatchNorm2d (state-dict compatible) and fuses
+the activation and residual add into the normalization pass, replacing the
+MIOpen BN kernel stack + separate relu/add elementwise kernels in the
+ResNet/DenseNet hot path (csrc/bn_kernels.hip; motivation in
+profiles/README.md).
+
+Fast path requires: CUDA + fp32 + channels_last.  Anything else falls back to
+a torch composition with IDENTICAL numerics (so models stay CPU-testable and
+eval/fp16 paths just work).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+__all__ = ["FusedBNAct2d"]
+
+
+def _nparts(rows: int, C: int) -> int:
+    cblocks = (C + 255) // 256
+    n = max(2048 // max(cblocks, 1), 1)
+    return int(min(n, rows, 1024))
+
+
+class _FusedBNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var, training,
+                momentum, eps, relu, residual):
+        import dear_pytorch_amd._kernels as K
+        N, C, H, W = x.shape
+        rows = N * H * W
+        y = torch.empty_like(x)
+        if training:
+            mean = torch.empty(C, device=x.device, dtype=torch.float32)
+            invstd = torch.empty_like(mean)
+        else:
+            mean = running_mean
+            invstd = torch.rsqrt(running_var + eps)
+        np_ = _nparts(rows, C)
+        psum = torch.empty(np_, C, device=x.device, dtype=torch.float32)
+        psumsq = torch.empty_like(psum)
+        K.bn_fwd(x, residual, y, weight, bias, mean, invstd,
+                 running_mean, running_var, psum, psumsq, rows, C,
+                 eps, momentum, training, relu)
+        ctx.save_for_backward(x, y, weight, mean, invstd)
+        ctx.relu = relu
+        ctx.has_res = residual is not None
+        ctx.dims = (rows, C)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        import dear_pytorch_amd._kernels as K
+        x, y, weight, mean, invstd = ctx.saved_tensors
+        rows, C = ctx.dims
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = torch.empty_like(x)
+        dgamma = torch.empty(C, device=x.device, dtype=torch.float32)
+        dbeta = torch.empty_like(dgamma)
+        np_ = _nparts(rows, C)
+        pdb = torch.empty(np_, C, device=x.device, dtype=torch.float32)
+        pdg = torch.empty_like(pdb)
+        dy_eff = torch.empty_like(x) if ctx.has_res else None
+        K.bn_bwd(x, dy, y, dy_eff, weight, mean, invstd, pdb, pdg,
+                 dbeta, dgamma, dx, rows, C, ctx.relu, ctx.has_res)
+        d_res = dy_eff if ctx.has_res else None
+        return (dx, dgamma, dbeta, None, None, None, None, None, None, d_res)
+
+
+class FusedBNAct2d(nn.BatchNorm2d):
+    """BatchNorm2d with optional fused ReLU and fused residual add.
+
+    forward(x, residual=None) computes
+        act(bn(x) + residual)   with act = ReLU if relu else identity
+    """
+
+    def __init__(self, num_features, relu=False, **kw):
+        super().__init__(num_features, **kw)
+        self.relu = relu
+
+    def _fast_ok(self, x, residual):
+        return (x.is_cuda and x.dtype == torch.float32
+                and self.affine and self.track_running_stats
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and (residual is None or
+                     residual.is_contiguous(memory_format=torch.channels_last)))
+
+    def forward(self, x, residual: Optional[torch.Tensor] = None):
+        if self._fast_ok(x, residual):
+            if self.training and self.num_batches_tracked is not None:
+                self.num_batches_tracked += 1
+            mom = self.momentum if self.momentum is not None else 0.1
+            return _FusedBNFn.apply(x, self.weight, self.bias,
+                                    self.running_mean, self.running_var,
+                                    self.training, mom, self.eps, self.relu,
+                                    residual)
+        # reference-numerics fallback (CPU, fp16, NCHW, no-affine...)
+        y = super().forward(x)
+        if residual is not None:
+            y = y + residual
+        return F.relu(y) if self.relu else y

@@ -34,7 +34,8 @@ ext_modules = [
     ),
     CUDAExtension(
         name="dear_pytorch_amd._kernels",
-        sources=["dear_pytorch_amd/csrc/kernels.hip"],
+        sources=["dear_pytorch_amd/csrc/kernels.hip",
+                 "dear_pytorch_amd/csrc/bn_kernels.hip"],
         libraries=["amdhip64"],
         **common,
     ),
