@@ -357,6 +357,7 @@ static void select_ge(at::Tensor x, double thr, at::Tensor out_idx,
 }
 
 // ---------------------------------------------------------- fused BN (NHWC)
+extern "C" int dear_bn_nparts(long, int);
 extern "C" void dear_bn_fwd(hipStream_t, const float*, const float*, float*,
                             const float*, const float*, float*, float*,
                             float*, float*, float*, float*, int, long, int,
@@ -404,6 +405,9 @@ static void bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "DeAR CDNA4 fused kernels (gfx950)";
+  m.def("bn_nparts", [](int64_t rows, int64_t C) {
+    return dear_bn_nparts((long)rows, (int)C);
+  });
   m.def("bn_fwd", &bn_fwd);
   m.def("bn_bwd", &bn_bwd);
   m.def("fused_sgd", &fused_sgd);

@@ -22,9 +22,8 @@ __all__ = ["FusedBNAct2d"]
 
 
 def _nparts(rows: int, C: int) -> int:
-    cblocks = (C + 255) // 256
-    n = max(2048 // max(cblocks, 1), 1)
-    return int(min(n, rows, 1024))
+    import dear_pytorch_amd._kernels as K
+    return int(K.bn_nparts(rows, C))
 
 
 class _FusedBNFn(torch.autograd.Function):
@@ -85,6 +84,7 @@ class FusedBNAct2d(nn.BatchNorm2d):
 
     def _fast_ok(self, x, residual):
         return (x.is_cuda and x.dtype == torch.float32
+                and x.shape[1] % 4 == 0
                 and self.affine and self.track_running_stats
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and (residual is None or
