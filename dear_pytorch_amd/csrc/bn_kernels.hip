@@ -243,8 +243,9 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_dx_kernel(
 
 struct LaunchCfg {
   Geom g;
-  int rb;       // row blocks (gridDim.x)
-  int cblocks;  // gridDim.y
+  int rb;        // row blocks for the elementwise kernels (gridDim.x)
+  int stats_rb;  // row blocks for the reduction kernels (= #partials)
+  int cblocks;   // gridDim.y
 };
 
 LaunchCfg make_cfg(long rows, int C4, int nparts_cap) {
@@ -252,12 +253,24 @@ LaunchCfg make_cfg(long rows, int C4, int nparts_cap) {
   cfg.g.cpb = C4 < kThreads ? C4 : kThreads;
   cfg.g.rpb = kThreads / cfg.g.cpb;
   cfg.cblocks = (C4 + cfg.g.cpb - 1) / cfg.g.cpb;
-  long target = 1024 / cfg.cblocks;  // aim ~1024 workgroups total
-  if (target < 1) target = 1;
   long max_rb = (rows + cfg.g.rpb - 1) / cfg.g.rpb;
+  // elementwise kernels: fill the chip, grid-stride the rest
+  long target = 2048 / cfg.cblocks;
+  if (target < 1) target = 1;
   cfg.rb = (int)(target < max_rb ? target : max_rb);
   if (cfg.rb < 1) cfg.rb = 1;
-  if (nparts_cap > 0 && cfg.rb > nparts_cap) cfg.rb = nparts_cap;
+  // reduction kernels: each block should chew >= 32 row-iterations so the
+  // partial buffers stay small vs the tensor (v2 used a fixed 1024 and the
+  // partial traffic dominated small layers)
+  long srb = (max_rb + 31) / 32;
+  long scap = 1024 / cfg.cblocks;
+  if (scap < 8) scap = 8;
+  if (srb > scap) srb = scap;
+  if (srb > max_rb) srb = max_rb;
+  if (srb < 1) srb = 1;
+  cfg.stats_rb = (int)srb;
+  if (nparts_cap > 0 && cfg.stats_rb > nparts_cap)
+    cfg.stats_rb = nparts_cap;
   return cfg;
 }
 
@@ -267,7 +280,7 @@ extern "C" {
 
 // host-visible helper so the python side can size partial buffers identically
 int dear_bn_nparts(long rows, int C) {
-  return make_cfg(rows, C / 4, 0).rb;
+  return make_cfg(rows, C / 4, 0).stats_rb;
 }
 
 void dear_bn_fwd(hipStream_t stream, const float* x, const float* res,
@@ -278,13 +291,14 @@ void dear_bn_fwd(hipStream_t stream, const float* x, const float* res,
   const int C4 = C / 4;
   LaunchCfg cfg = make_cfg(rows, C4, nparts);
   if (training) {
-    hipLaunchKernelGGL(bn_fwd_stats_kernel, dim3(cfg.rb, cfg.cblocks),
-                       dim3(kThreads), 0, stream, (const float4*)x, rows, C4,
-                       cfg.g, (float4*)psum, (float4*)psumsq);
+    hipLaunchKernelGGL(bn_fwd_stats_kernel,
+                       dim3(cfg.stats_rb, cfg.cblocks), dim3(kThreads), 0,
+                       stream, (const float4*)x, rows, C4, cfg.g,
+                       (float4*)psum, (float4*)psumsq);
     const int crb = (C + kThreads - 1) / kThreads;
     hipLaunchKernelGGL(bn_fwd_reduce_kernel, dim3(crb), dim3(kThreads), 0,
-                       stream, psum, psumsq, cfg.rb, C, rows, eps, momentum,
-                       mean, invstd, running_mean, running_var, 1);
+                       stream, psum, psumsq, cfg.stats_rb, C, rows, eps,
+                       momentum, mean, invstd, running_mean, running_var, 1);
   }
 #define APPLY(R, S)                                                           \
   hipLaunchKernelGGL((bn_fwd_apply_kernel<R, S>),                             \
@@ -308,7 +322,8 @@ void dear_bn_bwd(hipStream_t stream, const float* x, const float* dy,
   LaunchCfg cfg = make_cfg(rows, C4, nparts);
 #define STATS(R, S)                                                           \
   hipLaunchKernelGGL((bn_bwd_stats_kernel<R, S>),                             \
-                     dim3(cfg.rb, cfg.cblocks), dim3(kThreads), 0, stream,    \
+                     dim3(cfg.stats_rb, cfg.cblocks), dim3(kThreads), 0,      \
+                     stream,                                                  \
                      (const float4*)x, (const float4*)dy, (const float4*)y,   \
                      (float4*)dy_eff, (const float4*)mean,                    \
                      (const float4*)invstd, rows, C4, cfg.g, (float4*)pdb,    \
@@ -320,7 +335,7 @@ void dear_bn_bwd(hipStream_t stream, const float* x, const float* dy,
 #undef STATS
   const int crb = (C + kThreads - 1) / kThreads;
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(crb), dim3(kThreads), 0,
-                     stream, pdb, pdg, cfg.rb, C, dbeta, dgamma);
+                     stream, pdb, pdg, cfg.stats_rb, C, dbeta, dgamma);
 #define DX(R, H)                                                              \
   hipLaunchKernelGGL((bn_bwd_dx_kernel<R, H>), dim3(cfg.rb, cfg.cblocks),     \
                      dim3(kThreads), 0, stream, (const float4*)x,             \

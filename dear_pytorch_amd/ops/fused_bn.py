@@ -26,30 +26,44 @@ def _nparts(rows: int, C: int) -> int:
     return int(K.bn_nparts(rows, C))
 
 
+class _Workspace:
+    """Per-(module, input-shape) scratch reused every iteration: stats
+    partials + mean/invstd.  Gradients are NOT cached (autograd may adopt the
+    returned tensors as .grad)."""
+
+    __slots__ = ("mean", "invstd", "psum", "psumsq", "nparts", "rows")
+
+    def __init__(self, rows, C, device):
+        self.rows = rows
+        self.nparts = _nparts(rows, C)
+        self.mean = torch.empty(C, device=device, dtype=torch.float32)
+        self.invstd = torch.empty_like(self.mean)
+        self.psum = torch.empty(self.nparts, C, device=device,
+                                dtype=torch.float32)
+        self.psumsq = torch.empty_like(self.psum)
+
+
 class _FusedBNFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, training,
-                momentum, eps, relu, residual):
+                momentum, eps, relu, residual, ws):
         import dear_pytorch_amd._kernels as K
         N, C, H, W = x.shape
         rows = N * H * W
         y = torch.empty_like(x)
         if training:
-            mean = torch.empty(C, device=x.device, dtype=torch.float32)
-            invstd = torch.empty_like(mean)
+            mean, invstd = ws.mean, ws.invstd
         else:
             mean = running_mean
             invstd = torch.rsqrt(running_var + eps)
-        np_ = _nparts(rows, C)
-        psum = torch.empty(np_, C, device=x.device, dtype=torch.float32)
-        psumsq = torch.empty_like(psum)
         K.bn_fwd(x, residual, y, weight, bias, mean, invstd,
-                 running_mean, running_var, psum, psumsq, rows, C,
+                 running_mean, running_var, ws.psum, ws.psumsq, rows, C,
                  eps, momentum, training, relu)
         ctx.save_for_backward(x, y, weight, mean, invstd)
         ctx.relu = relu
         ctx.has_res = residual is not None
         ctx.dims = (rows, C)
+        ctx.ws = ws
         return y
 
     @staticmethod
@@ -57,18 +71,18 @@ class _FusedBNFn(torch.autograd.Function):
         import dear_pytorch_amd._kernels as K
         x, y, weight, mean, invstd = ctx.saved_tensors
         rows, C = ctx.dims
+        ws = ctx.ws
         dy = dy.contiguous(memory_format=torch.channels_last)
         dx = torch.empty_like(x)
         dgamma = torch.empty(C, device=x.device, dtype=torch.float32)
         dbeta = torch.empty_like(dgamma)
-        np_ = _nparts(rows, C)
-        pdb = torch.empty(np_, C, device=x.device, dtype=torch.float32)
-        pdg = torch.empty_like(pdb)
         dy_eff = torch.empty_like(x) if ctx.has_res else None
-        K.bn_bwd(x, dy, y, dy_eff, weight, mean, invstd, pdb, pdg,
+        # partial buffers reused from forward (forward's reduce consumed them)
+        K.bn_bwd(x, dy, y, dy_eff, weight, mean, invstd, ws.psum, ws.psumsq,
                  dbeta, dgamma, dx, rows, C, ctx.relu, ctx.has_res)
         d_res = dy_eff if ctx.has_res else None
-        return (dx, dgamma, dbeta, None, None, None, None, None, None, d_res)
+        return (dx, dgamma, dbeta, None, None, None, None, None, None, d_res,
+                None)
 
 
 class FusedBNAct2d(nn.BatchNorm2d):
@@ -81,6 +95,7 @@ class FusedBNAct2d(nn.BatchNorm2d):
     def __init__(self, num_features, relu=False, **kw):
         super().__init__(num_features, **kw)
         self.relu = relu
+        self._ws = None
 
     def _fast_ok(self, x, residual):
         return (x.is_cuda and x.dtype == torch.float32
@@ -95,10 +110,15 @@ class FusedBNAct2d(nn.BatchNorm2d):
             if self.training and self.num_batches_tracked is not None:
                 self.num_batches_tracked += 1
             mom = self.momentum if self.momentum is not None else 0.1
+            N, C, H, W = x.shape
+            rows = N * H * W
+            if self._ws is None or self._ws.rows != rows \
+                    or self._ws.mean.device != x.device:
+                self._ws = _Workspace(rows, C, x.device)
             return _FusedBNFn.apply(x, self.weight, self.bias,
                                     self.running_mean, self.running_var,
                                     self.training, mom, self.eps, self.relu,
-                                    residual)
+                                    residual, self._ws)
         # reference-numerics fallback (CPU, fp16, NCHW, no-affine...)
         y = super().forward(x)
         if residual is not None:
