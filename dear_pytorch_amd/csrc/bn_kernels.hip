@@ -39,6 +39,16 @@ __device__ __forceinline__ bool lane_map(const Geom g, int C4, int& c4,
   return (c4 < C4) && (rsub < g.rpb);
 }
 
+// contiguous slab of rows for this block (DRAM-page locality beats a
+// whole-grid stride; measured on the fused_sgd chunks at ~5.8 TB/s)
+__device__ __forceinline__ void row_range(long rows, int rpb, long& r0,
+                                          long& r1) {
+  const long per = ((rows + (long)gridDim.x * rpb - 1) /
+                    ((long)gridDim.x * rpb)) * rpb;
+  r0 = (long)blockIdx.x * per;
+  r1 = r0 + per < rows ? r0 + per : rows;
+}
+
 // ---- forward statistics: partial per-channel sum / sumsq ------------------
 __global__ __launch_bounds__(kThreads) void bn_fwd_stats_kernel(
     const float4* __restrict__ x, long rows, int C4, Geom g,
@@ -48,14 +58,24 @@ __global__ __launch_bounds__(kThreads) void bn_fwd_stats_kernel(
   const bool act = lane_map(g, C4, c4, rsub);
   float4 s = make_float4(0, 0, 0, 0), ss = make_float4(0, 0, 0, 0);
   if (act) {
-    const long step = (long)gridDim.x * g.rpb;
-    for (long r = (long)blockIdx.x * g.rpb + rsub; r < rows; r += step) {
+    long r0, r1;
+    row_range(rows, g.rpb, r0, r1);
+    long r = r0 + rsub;
+    for (; r + g.rpb < r1; r += 2 * g.rpb) {
+      float4 v0 = x[r * C4 + c4];
+      float4 v1 = x[(r + g.rpb) * C4 + c4];
+      s = s + v0;
+      ss.x = fmaf(v0.x, v0.x, ss.x); ss.y = fmaf(v0.y, v0.y, ss.y);
+      ss.z = fmaf(v0.z, v0.z, ss.z); ss.w = fmaf(v0.w, v0.w, ss.w);
+      s = s + v1;
+      ss.x = fmaf(v1.x, v1.x, ss.x); ss.y = fmaf(v1.y, v1.y, ss.y);
+      ss.z = fmaf(v1.z, v1.z, ss.z); ss.w = fmaf(v1.w, v1.w, ss.w);
+    }
+    if (r < r1) {
       float4 v = x[r * C4 + c4];
       s = s + v;
-      ss.x = fmaf(v.x, v.x, ss.x);
-      ss.y = fmaf(v.y, v.y, ss.y);
-      ss.z = fmaf(v.z, v.z, ss.z);
-      ss.w = fmaf(v.w, v.w, ss.w);
+      ss.x = fmaf(v.x, v.x, ss.x); ss.y = fmaf(v.y, v.y, ss.y);
+      ss.z = fmaf(v.z, v.z, ss.z); ss.w = fmaf(v.w, v.w, ss.w);
     }
   }
   ls[threadIdx.x] = s;
@@ -114,9 +134,9 @@ __global__ __launch_bounds__(kThreads) void bn_fwd_apply_kernel(
   scale.y = ic.y * wc.y; shift.y = fmaf(-mc.y, scale.y, bc.y);
   scale.z = ic.z * wc.z; shift.z = fmaf(-mc.z, scale.z, bc.z);
   scale.w = ic.w * wc.w; shift.w = fmaf(-mc.w, scale.w, bc.w);
-  const long step = (long)gridDim.x * g.rpb;
-  for (long r = (long)blockIdx.x * g.rpb + rsub; r < rows; r += step) {
-    const long i = r * C4 + c4;
+  long r0, r1;
+  row_range(rows, g.rpb, r0, r1);
+  auto body = [&](long i) {
     float4 v = x[i];
     v.x = fmaf(v.x, scale.x, shift.x);
     v.y = fmaf(v.y, scale.y, shift.y);
@@ -131,7 +151,13 @@ __global__ __launch_bounds__(kThreads) void bn_fwd_apply_kernel(
       v.z = fmaxf(v.z, 0.f); v.w = fmaxf(v.w, 0.f);
     }
     y[i] = v;
+  };
+  long r = r0 + rsub;
+  for (; r + g.rpb < r1; r += 2 * g.rpb) {
+    body(r * C4 + c4);
+    body((r + g.rpb) * C4 + c4);
   }
+  if (r < r1) body(r * C4 + c4);
 }
 
 // ---- backward statistics: dy_eff, partial Σdy_eff and Σdy_eff·x̂ -----------
@@ -148,8 +174,9 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_stats_kernel(
   float4 sdb = make_float4(0, 0, 0, 0), sdg = make_float4(0, 0, 0, 0);
   if (act) {
     const float4 mc = mean[c4], ic = invstd[c4];
-    const long step = (long)gridDim.x * g.rpb;
-    for (long r = (long)blockIdx.x * g.rpb + rsub; r < rows; r += step) {
+    long r0, r1;
+    row_range(rows, g.rpb, r0, r1);
+    for (long r = r0 + rsub; r < r1; r += g.rpb) {
       const long i = r * C4 + c4;
       float4 gg = dy[i];
       if (kRelu) {
@@ -215,8 +242,9 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_dx_kernel(
   k.y = ic.y * wc.y; mdb.y = db.y * inv_m; mdg.y = dg.y * inv_m;
   k.z = ic.z * wc.z; mdb.z = db.z * inv_m; mdg.z = dg.z * inv_m;
   k.w = ic.w * wc.w; mdb.w = db.w * inv_m; mdg.w = dg.w * inv_m;
-  const long step = (long)gridDim.x * g.rpb;
-  for (long r = (long)blockIdx.x * g.rpb + rsub; r < rows; r += step) {
+  long r0, r1;
+  row_range(rows, g.rpb, r0, r1);
+  for (long r = r0 + rsub; r < r1; r += g.rpb) {
     const long i = r * C4 + c4;
     float4 gg;
     if (kHaveDyEff) {
