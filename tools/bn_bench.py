@@ -1,6 +1,7 @@
 #!/usr/bin/env python
 """Microbench: FusedBNAct2d vs nn.BatchNorm2d(+relu)(+add) on ResNet-50
 shapes (NHWC fp32). Prints per-op fwd / fwd+bwd times."""
+import os
 import sys
 import time
 
@@ -20,7 +21,7 @@ SHAPES = [  # (N, C, H, W, res?)
 ]
 
 
-def timeit(fn, iters=20, warmup=5):
+def timeit(fn, iters=int(os.environ.get("BN_ITERS","20")), warmup=int(os.environ.get("BN_WARMUP","5")) ):
     for _ in range(warmup):
         fn()
     torch.cuda.synchronize()
