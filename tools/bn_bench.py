@@ -9,7 +9,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from dear_pytorch_amd.ops.fused_bn import FusedBNAct2d  # noqa: E402
 
 SHAPES = [  # (N, C, H, W, res?)
