@@ -92,30 +92,53 @@ __global__ __launch_bounds__(kThreads) void bn_fwd_stats_kernel(
   }
 }
 
-// ---- finalize mean / invstd + running stats (scalar over C, tiny) ---------
+// ---- finalize mean / invstd + running stats -------------------------------
+// Cooperative column reduce of the [nparts][C4] float4 partial matrix:
+// a block covers cpb2 float4-columns x spb partial-slices (the v2 scalar
+// one-thread-per-channel loop measured 179 us -- it was the whole overhead).
 __global__ __launch_bounds__(kThreads) void bn_fwd_reduce_kernel(
-    const float* __restrict__ psum, const float* __restrict__ psumsq,
-    int nparts, int C, long rows, float eps, float momentum,
+    const float4* __restrict__ psum, const float4* __restrict__ psumsq,
+    int nparts, int C4, long rows, float eps, float momentum,
     float* __restrict__ mean, float* __restrict__ invstd,
     float* __restrict__ running_mean, float* __restrict__ running_var,
-    int update_running) {
-  const int c = blockIdx.x * kThreads + threadIdx.x;
-  if (c >= C) return;
-  float s = 0.f, ss = 0.f;
-  for (int p = 0; p < nparts; ++p) {
-    s += psum[(long)p * C + c];
-    ss += psumsq[(long)p * C + c];
+    int update_running, int cpb2) {
+  __shared__ float4 l1[kThreads], l2[kThreads];
+  const int spb = kThreads / cpb2;
+  const int cl = threadIdx.x % cpb2;
+  const int sl = threadIdx.x / cpb2;
+  const int c4 = blockIdx.x * cpb2 + cl;
+  float4 s = make_float4(0, 0, 0, 0), ss = make_float4(0, 0, 0, 0);
+  if (c4 < C4 && sl < spb) {
+    for (int p = sl; p < nparts; p += spb) {
+      s = s + psum[(long)p * C4 + c4];
+      ss = ss + psumsq[(long)p * C4 + c4];
+    }
   }
-  const float m = s / (float)rows;
-  float var = fmaxf(ss / (float)rows - m * m, 0.f);
-  mean[c] = m;
-  invstd[c] = rsqrtf(var + eps);
-  if (update_running) {
-    const float unbiased = rows > 1 ? var * (float)rows / (float)(rows - 1)
-                                    : var;
-    running_mean[c] = fmaf(momentum, m - running_mean[c], running_mean[c]);
-    running_var[c] = fmaf(momentum, unbiased - running_var[c],
-                          running_var[c]);
+  l1[threadIdx.x] = s;
+  l2[threadIdx.x] = ss;
+  __syncthreads();
+  if (sl == 0 && c4 < C4) {
+    for (int k = 1; k < spb; ++k) {
+      s = s + l1[k * cpb2 + cl];
+      ss = ss + l2[k * cpb2 + cl];
+    }
+    const float inv_rows = 1.f / (float)rows;
+    const float unb = rows > 1 ? (float)rows / (float)(rows - 1) : 1.f;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int c = c4 * 4 + j;
+      const float sj = (&s.x)[j], ssj = (&ss.x)[j];
+      const float m = sj * inv_rows;
+      float var = fmaxf(ssj * inv_rows - m * m, 0.f);
+      mean[c] = m;
+      invstd[c] = rsqrtf(var + eps);
+      if (update_running) {
+        running_mean[c] = fmaf(momentum, m - running_mean[c],
+                               running_mean[c]);
+        running_var[c] = fmaf(momentum, var * unb - running_var[c],
+                              running_var[c]);
+      }
+    }
   }
 }
 
@@ -210,17 +233,35 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_stats_kernel(
 }
 
 __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_kernel(
-    const float* __restrict__ pdb, const float* __restrict__ pdg, int nparts,
-    int C, float* __restrict__ dbeta, float* __restrict__ dgamma) {
-  const int c = blockIdx.x * kThreads + threadIdx.x;
-  if (c >= C) return;
-  float db = 0.f, dg = 0.f;
-  for (int p = 0; p < nparts; ++p) {
-    db += pdb[(long)p * C + c];
-    dg += pdg[(long)p * C + c];
+    const float4* __restrict__ pdb, const float4* __restrict__ pdg,
+    int nparts, int C4, float* __restrict__ dbeta, float* __restrict__ dgamma,
+    int cpb2) {
+  __shared__ float4 l1[kThreads], l2[kThreads];
+  const int spb = kThreads / cpb2;
+  const int cl = threadIdx.x % cpb2;
+  const int sl = threadIdx.x / cpb2;
+  const int c4 = blockIdx.x * cpb2 + cl;
+  float4 db = make_float4(0, 0, 0, 0), dg = make_float4(0, 0, 0, 0);
+  if (c4 < C4 && sl < spb) {
+    for (int p = sl; p < nparts; p += spb) {
+      db = db + pdb[(long)p * C4 + c4];
+      dg = dg + pdg[(long)p * C4 + c4];
+    }
   }
-  dbeta[c] = db;
-  dgamma[c] = dg;
+  l1[threadIdx.x] = db;
+  l2[threadIdx.x] = dg;
+  __syncthreads();
+  if (sl == 0 && c4 < C4) {
+    for (int k = 1; k < spb; ++k) {
+      db = db + l1[k * cpb2 + cl];
+      dg = dg + l2[k * cpb2 + cl];
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      dbeta[c4 * 4 + j] = (&db.x)[j];
+      dgamma[c4 * 4 + j] = (&dg.x)[j];
+    }
+  }
 }
 
 // ---- backward dx: dx = w*invstd*(dy_eff - Σdb/M - x̂·Σdg/M) ----------------
@@ -323,10 +364,12 @@ void dear_bn_fwd(hipStream_t stream, const float* x, const float* res,
                        dim3(cfg.stats_rb, cfg.cblocks), dim3(kThreads), 0,
                        stream, (const float4*)x, rows, C4, cfg.g,
                        (float4*)psum, (float4*)psumsq);
-    const int crb = (C + kThreads - 1) / kThreads;
+    const int cpb2 = C4 < 64 ? C4 : 64;
+    const int crb = (C4 + cpb2 - 1) / cpb2;
     hipLaunchKernelGGL(bn_fwd_reduce_kernel, dim3(crb), dim3(kThreads), 0,
-                       stream, psum, psumsq, cfg.stats_rb, C, rows, eps,
-                       momentum, mean, invstd, running_mean, running_var, 1);
+                       stream, (const float4*)psum, (const float4*)psumsq,
+                       cfg.stats_rb, C4, rows, eps, momentum, mean, invstd,
+                       running_mean, running_var, 1, cpb2);
   }
 #define APPLY(R, S)                                                           \
   hipLaunchKernelGGL((bn_fwd_apply_kernel<R, S>),                             \
@@ -361,9 +404,11 @@ void dear_bn_bwd(hipStream_t stream, const float* x, const float* dy,
   else if (want_dy_eff) STATS(false, true);
   else STATS(false, false);
 #undef STATS
-  const int crb = (C + kThreads - 1) / kThreads;
+  const int cpb2 = C4 < 64 ? C4 : 64;
+  const int crb = (C4 + cpb2 - 1) / cpb2;
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(crb), dim3(kThreads), 0,
-                     stream, pdb, pdg, cfg.stats_rb, C, dbeta, dgamma);
+                     stream, (const float4*)pdb, (const float4*)pdg,
+                     cfg.stats_rb, C4, dbeta, dgamma, cpb2);
 #define DX(R, H)                                                              \
   hipLaunchKernelGGL((bn_bwd_dx_kernel<R, H>), dim3(cfg.rb, cfg.cblocks),     \
                      dim3(kThreads), 0, stream, (const float4*)x,             \
