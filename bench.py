@@ -42,6 +42,9 @@ def parse_args():
     p.add_argument("--comm-dtype", default="fp32",
                    choices=["fp32", "bf16", "fp16"],
                    help="gradient wire format (bf16/fp16 halve xGMI bytes)")
+    p.add_argument("--hipgraph", action="store_true",
+                   help="capture the whole training step in a hipGraph and "
+                        "replay it (removes per-kernel launch gaps)")
     p.add_argument("--fused-bn", action="store_true", default=None,
                    help="fused NHWC BN+ReLU+residual kernels for ResNets "
                         "(default on for GPU ResNets)")
@@ -158,12 +161,29 @@ def main():
         if on_gpu:
             torch.cuda.synchronize()
 
+    run_step = lambda: step_fn(model, opt)
+    if args.hipgraph and on_gpu:
+        # warm up allocator/find on a side stream, then capture one full
+        # iteration (fwd + bwd + DeAR update enqueues — all device-static:
+        # fused buckets, grad views and synthetic inputs never reallocate)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                step_fn(model, opt)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            step_fn(model, opt)
+        run_step = graph.replay
+
     for _ in range(args.warmup):
-        step_fn(model, opt)
+        run_step()
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step_fn(model, opt)
+        run_step()
     sync()
     elapsed = time.perf_counter() - t0
     # MAX over ranks
