@@ -152,7 +152,15 @@ def main():
     model, opt_fn, step_fn, bs, unit, metric = build_workload(args, device)
     if world > 1:
         dear.broadcast_parameters(model.state_dict(), root_rank=0)
-    model, opt = wrap_method(args, model, opt_fn)
+    # hipGraph mode: hook/AccumulateGrad creation, warmup and capture must all
+    # live on ONE non-default stream or backward breaks the capture
+    graph_stream = torch.cuda.Stream() if (args.hipgraph and on_gpu) else None
+    if graph_stream is not None:
+        graph_stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(graph_stream):
+            model, opt = wrap_method(args, model, opt_fn)
+    else:
+        model, opt = wrap_method(args, model, opt_fn)
     model.train()
 
     def sync():
@@ -162,19 +170,17 @@ def main():
             torch.cuda.synchronize()
 
     run_step = lambda: step_fn(model, opt)
-    if args.hipgraph and on_gpu:
-        # warm up allocator/find on a side stream, then capture one full
-        # iteration (fwd + bwd + DeAR update enqueues — all device-static:
-        # fused buckets, grad views and synthetic inputs never reallocate)
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):
+    if graph_stream is not None:
+        # warm up allocator/find, then capture one full iteration (fwd + bwd
+        # + DeAR update enqueues — all device-static: fused buckets, grad
+        # views and synthetic inputs never reallocate), all on graph_stream
+        with torch.cuda.stream(graph_stream):
             for _ in range(3):
                 step_fn(model, opt)
-        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.current_stream().wait_stream(graph_stream)
         torch.cuda.synchronize()
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        with torch.cuda.graph(graph, stream=graph_stream):
             step_fn(model, opt)
         run_step = graph.replay
 
