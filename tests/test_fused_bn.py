@@ -112,3 +112,39 @@ def test_gpu_eval_mode_matches():
         y1 = fused(x)
         y2 = F.relu(ref_bn(x))
     assert torch.allclose(y1, y2, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_gpu_fused_resnet50_e2e_matches_plain():
+    """Full-model check: fused-BN ResNet-50 tracks the plain model's loss
+    trajectory and parameters over 3 DeAR training steps (fp32, NHWC)."""
+    import dear_pytorch_amd as dear
+    from dear_pytorch_amd import models
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    a = models.get_cnn("resnet50", num_classes=100, fused_bn=False).to(dev)
+    b = models.get_cnn("resnet50", num_classes=100, fused_bn=True).to(dev)
+    b.load_state_dict(a.state_dict())
+    a = a.to(memory_format=torch.channels_last)
+    b = b.to(memory_format=torch.channels_last)
+    x = torch.randn(8, 3, 224, 224, device=dev) \
+        .to(memory_format=torch.channels_last)
+    y = torch.randint(0, 100, (8,), device=dev)
+    lossf = nn.CrossEntropyLoss()
+    oa = dear.DistributedOptimizer(
+        torch.optim.SGD(a.parameters(), lr=0.01, momentum=0.9), model=a)
+    ob = dear.DistributedOptimizer(
+        torch.optim.SGD(b.parameters(), lr=0.01, momentum=0.9), model=b)
+    for i in range(3):
+        la = lossf(a(x), y)
+        la.backward()
+        oa.step()
+        lb = lossf(b(x), y)
+        lb.backward()
+        ob.step()
+        assert abs(la.item() - lb.item()) < 5e-3 * max(1.0, la.item()), \
+            (i, la.item(), lb.item())
+    oa.synchronize(); ob.synchronize()
+    worst = max((pa - pb).abs().max().item()
+                for pa, pb in zip(a.parameters(), b.parameters()))
+    assert worst < 5e-3, worst
