@@ -38,6 +38,9 @@ def main():
     p.add_argument("--compressor", default="none")
     p.add_argument("--density", type=float, default=1.0)
     p.add_argument("--timeline", default="")
+    p.add_argument("--amp", action="store_true",
+                   help="bf16 autocast compute (reference --fp16 axis; "
+                        "gradients/optimizer stay fp32)")
     args = p.parse_args()
 
     import dear_pytorch_amd as dear
@@ -113,11 +116,15 @@ def main():
 
     model.train()
 
+    amp_ctx = (lambda: torch.autocast("cuda", dtype=torch.bfloat16)) \
+        if (args.amp and on_gpu) else torch.enable_grad
+
     def benchmark_step():
         if tuner:
             tuner.step_begin()
         opt.zero_grad()
-        loss = lossf(model(data), target)
+        with amp_ctx():
+            loss = lossf(model(data), target)
         loss.backward()
         opt.step()
         if tuner:
