@@ -30,6 +30,9 @@ def main():
     p.add_argument("--no-fusion", action="store_true")
     p.add_argument("--exclude-parts", default="")
     p.add_argument("--optimizer", default="sgd", choices=["sgd", "adamw"])
+    p.add_argument("--amp", action="store_true",
+                   help="bf16 autocast compute (reference --fp16 axis; "
+                        "gradients/optimizer stay fp32)")
     args = p.parse_args()
 
     import dear_pytorch_amd as dear
@@ -88,12 +91,16 @@ def main():
 
     model.train()
 
+    amp_ctx = (lambda: torch.autocast("cuda", dtype=torch.bfloat16)) \
+        if (args.amp and on_gpu) else torch.enable_grad
+
     def benchmark_step():
         if tuner:
             tuner.step_begin()
         opt.zero_grad()
-        scores, seq_rel = model(ids, tt, mask)
-        loss = crit(scores, seq_rel, mlm, nsp)
+        with amp_ctx():
+            scores, seq_rel = model(ids, tt, mask)
+            loss = crit(scores, seq_rel, mlm, nsp)
         loss.backward()
         opt.step()
         if tuner:
