@@ -254,3 +254,94 @@ def test_bytescheduler_ws2_matches_serial():
     for k in ref:
         assert torch.allclose(ref[k], outs[0][k], atol=1e-5), k
         assert torch.equal(outs[0][k], outs[1][k])
+
+
+def _rank_bert_tied(rank, world):
+    import torch
+    import dear_pytorch_amd as dear
+    from dear_pytorch_amd import models
+    dear.init(backend="gloo")
+    torch.manual_seed(0)
+    cfg = models.bert_base()
+    cfg.num_hidden_layers = 1
+    cfg.hidden_size = 32
+    cfg.num_attention_heads = 2
+    cfg.intermediate_size = 64
+    cfg.vocab_size = 128
+    m = models.BertForPreTraining(cfg)
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    crit = models.BertPretrainingCriterion(cfg.vocab_size)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=1e-3), model=m,
+        threshold_bytes=1 << 12)
+    # tied decoder/embedding weight must occupy exactly one bucket slot
+    n_slots = sum(len(g.slots) for g in opt.groups)
+    n_params = len({id(p) for p in m.parameters()})
+    assert n_slots == n_params
+    g = torch.Generator().manual_seed(5 + rank)
+    for _ in range(3):
+        ids = torch.randint(0, cfg.vocab_size, (2, 12), generator=g)
+        tt = torch.zeros(2, 12, dtype=torch.long)
+        mlm = torch.full((2, 12), -1)
+        mlm[:, 2] = ids[:, 2]
+        nsp = torch.randint(0, 2, (2,), generator=g)
+        opt.zero_grad()
+        scores, rel = m(ids, tt)
+        crit(scores, rel, mlm, nsp).backward()
+        opt.step()
+    opt.synchronize()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_bert_tied_weights_dear_ws2():
+    outs = run_dist(_rank_bert_tied, world_size=2)
+    for k in outs[0]:
+        assert torch.equal(outs[0][k], outs[1][k]), k
+        assert torch.isfinite(outs[0][k]).all(), k
+
+
+def _rank_hetero_param_groups(rank, world):
+    import torch
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    # two param groups with different lr -> shadow-optimizer path
+    pgs = [{"params": [p for n, p in m.named_parameters() if "weight" in n],
+            "lr": 0.05},
+           {"params": [p for n, p in m.named_parameters() if "bias" in n],
+            "lr": 0.01}]
+    opt = dear.DistributedOptimizer(torch.optim.SGD(pgs, momentum=0.9),
+                                    model=m, threshold_bytes=1 << 12)
+    for x, y in _full_data(4, 8):
+        xs, ys = x[rank * 8:(rank + 1) * 8], y[rank * 8:(rank + 1) * 8]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    opt.synchronize()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_hetero_param_groups_ws2():
+    # serial reference with the same two param groups
+    m = _model()
+    pgs = [{"params": [p for n, p in m.named_parameters() if "weight" in n],
+            "lr": 0.05},
+           {"params": [p for n, p in m.named_parameters() if "bias" in n],
+            "lr": 0.01}]
+    opt = torch.optim.SGD(pgs, momentum=0.9)
+    for x, y in _full_data(4, 8):
+        opt.zero_grad()
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+    ref = {k: v.clone() for k, v in m.state_dict().items()}
+    outs = run_dist(_rank_hetero_param_groups, world_size=2)
+    for k in ref:
+        assert torch.allclose(ref[k], outs[0][k], atol=1e-5), k
+        assert torch.equal(outs[0][k], outs[1][k])
