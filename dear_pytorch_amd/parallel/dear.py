@@ -112,19 +112,33 @@ class DearOptimizer(torch.optim.Optimizer):
                   flush=True)
 
     def _register_hooks(self):
-        # per-parameter grad-accumulator hooks -> reduce-scatter on group-complete
+        # per-parameter grad-accumulator hooks -> reduce-scatter on
+        # group-complete.  Handles are kept so regroup() can remove them —
+        # duplicate hooks would double-count readiness and fire the RS with
+        # half a group's gradients.
         self._grad_accs = []
+        self._bw_hook_handles = getattr(self, "_bw_hook_handles", [])
         for g in self.groups:
             for s in g.slots:
                 p = s.param
                 tmp = p.expand_as(p)
                 grad_acc = tmp.grad_fn.next_functions[0][0]
-                grad_acc.register_hook(self._make_bw_hook(p))
+                self._bw_hook_handles.append(
+                    grad_acc.register_hook(self._make_bw_hook(p)))
                 self._grad_accs.append(grad_acc)
         # forward-pre hook on the FIRST module of each group: sync AG + lazy update
         for g in self.groups:
             h = g.modules[0].register_forward_pre_hook(self._make_fw_hook(g))
             self._hook_handles.append(h)
+
+    def _remove_hooks(self):
+        for h in self._hook_handles:
+            h.remove()
+        self._hook_handles.clear()
+        for h in getattr(self, "_bw_hook_handles", []):
+            h.remove()
+        self._bw_hook_handles = []
+        self._grad_accs = []
 
     # ------------------------------------------------------------------ hooks
     def _make_bw_hook(self, p):
@@ -208,18 +222,17 @@ class DearOptimizer(torch.optim.Optimizer):
         loss = None
         if closure is not None:
             loss = closure()
-        if self._num_steps > 0 or True:
-            # bound host run-ahead to one iteration: wait for the PREVIOUS
-            # iteration's last AG (its results were consumed by this forward)
-            self._prev_iter_done.host_wait()
-            last = NULL_HANDLE
-            for g in self.groups:
-                self._ag_handle[g.index] = self._enqueue_gather(g)
-                if self._ag_handle[g.index] is not NULL_HANDLE:
-                    last = self._ag_handle[g.index]
-                self._updated[g.index] = False
-            self._prev_iter_done = last
-            self._ready_count = [0] * len(self.groups)
+        # bound host run-ahead to one iteration: wait for the PREVIOUS
+        # iteration's last AG (its results were consumed by this forward)
+        self._prev_iter_done.host_wait()
+        last = NULL_HANDLE
+        for g in self.groups:
+            self._ag_handle[g.index] = self._enqueue_gather(g)
+            if self._ag_handle[g.index] is not NULL_HANDLE:
+                last = self._ag_handle[g.index]
+            self._updated[g.index] = False
+        self._prev_iter_done = last
+        self._ready_count = [0] * len(self.groups)
         self._num_steps += 1
         return loss
 
@@ -238,9 +251,7 @@ class DearOptimizer(torch.optim.Optimizer):
     # dopt_rsag_bo.py:148-171 window).
     def regroup(self, threshold_bytes: int = None, fusion_flags=None):
         self.synchronize()
-        for h in self._hook_handles:
-            h.remove()
-        self._hook_handles.clear()
+        self._remove_hooks()
         for g in self.groups:
             fused_ops.detach_group_state(self.optim, g)
             g.free()

@@ -59,10 +59,9 @@ class WaitTimeAdaptiveFusion:
             return hook
 
         opt._make_bw_hook = make_hook
-        # re-register so wrapped hooks are live (groups unchanged)
-        for h in opt._hook_handles:
-            h.remove()
-        opt._hook_handles.clear()
+        # re-register so wrapped hooks are live (groups unchanged); the old
+        # backward hooks MUST be removed or readiness double-counts
+        opt._remove_hooks()
         opt._register_hooks()
 
     def step_end(self):

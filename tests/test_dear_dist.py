@@ -345,3 +345,42 @@ def test_hetero_param_groups_ws2():
     for k in ref:
         assert torch.allclose(ref[k], outs[0][k], atol=1e-5), k
         assert torch.equal(outs[0][k], outs[1][k])
+
+
+def _rank_regroup_ws2(rank, world):
+    import torch
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m,
+        threshold_bytes=1 << 12)
+    data = _full_data(6, 8)
+    for i, (x, y) in enumerate(data):
+        xs, ys = x[rank * 8:(rank + 1) * 8], y[rank * 8:(rank + 1) * 8]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+        if i == 2:
+            # regression: regroup used to leave the old backward hooks
+            # registered -> readiness double-counted -> RS fired with HALF a
+            # group's gradients on multi-rank runs
+            opt.regroup(1 << 14)
+    opt.synchronize()
+    # exactly one backward hook per param must be live
+    assert len(opt._bw_hook_handles) == \
+        len({id(p) for p in m.parameters() if p.requires_grad})
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_regroup_midtraining_ws2_matches_serial():
+    ref = _serial_reference(6, 8)
+    outs = run_dist(_rank_regroup_ws2, world_size=2)
+    for k in ref:
+        assert torch.allclose(ref[k], outs[0][k], atol=1e-5), \
+            f"{k}: {(ref[k] - outs[0][k]).abs().max():.3e}"
+        assert torch.equal(outs[0][k], outs[1][k])
