@@ -268,10 +268,18 @@ def _shadow_step(optim, group, scale):
 
 
 # --------------------------------------------------------------------- native
+def _ensure_desc(group):
+    """(Re)build the static chunk table; param storages can move (e.g. a late
+    .to(memory_format=...)), which would make cached device pointers stale."""
+    p0 = group.slots[0].param.data_ptr()
+    if "desc" not in group.extra or group.extra.get("desc_p0") != p0:
+        group.extra["desc"] = _build_desc(group)
+        group.extra["desc_p0"] = p0
+
+
 def _native_sgd(group, hyp, scale, first_step):
     ext = _native()
-    if "desc" not in group.extra:
-        group.extra["desc"] = _build_desc(group)
+    _ensure_desc(group)
     mom_buf = group.extra.get("momentum")
     if mom_buf is None:
         mom_buf = group.bucket  # unused when momentum == 0
@@ -285,8 +293,7 @@ def _native_sgd(group, hyp, scale, first_step):
 
 def _native_adam(group, hyp, scale, decoupled_wd, step):
     ext = _native()
-    if "desc" not in group.extra:
-        group.extra["desc"] = _build_desc(group)
+    _ensure_desc(group)
     b1, b2 = hyp.get("betas", (0.9, 0.999))
     ext.fused_adam(group.extra["desc"], group.bucket,
                    group.extra["exp_avg"], group.extra["exp_avg_sq"],
