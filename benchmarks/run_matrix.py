@@ -26,8 +26,11 @@ TASKS = [
 METHOD_SETS = {
     # with tensor fusion vs without (reference tf/notf sets)
     "tf": ["ddp", "mgwfbp", "dear"],
-    "notf": ["wfbp", "naive", "dear-notf"],
-    "all": ["ddp", "wfbp", "mgwfbp", "naive", "rb", "dear", "dear-bo"],
+    "notf": ["wfbp", "naive", "dear-notf", "bytescheduler"],
+    "all": ["ddp", "wfbp", "mgwfbp", "naive", "rb", "bytescheduler", "dear",
+            "dear-bo"],
+    # time-breakdown ablations (reference dear/batch.sh exclude_parts sweeps)
+    "breakdown": ["dear", "dear-nors", "dear-noag"],
 }
 
 
@@ -41,6 +44,10 @@ def gen_cmd(driver, model, bs, method, gpus, iters):
            "--num-iters", str(iters)]
     if method == "dear-notf":
         cmd += ["--method", "dear", "--no-fusion"]
+    elif method == "dear-nors":
+        cmd += ["--method", "dear", "--exclude-parts", "reducescatter"]
+    elif method == "dear-noag":
+        cmd += ["--method", "dear", "--exclude-parts", "allgather"]
     elif method == "wfbp":
         cmd += ["--method", "wfbp", "--no-fusion"]
     else:
@@ -66,6 +73,8 @@ def main():
     p.add_argument("--log", default="exp.log")
     p.add_argument("--out", default="reports.json")
     p.add_argument("--timeout", type=int, default=1800)
+    p.add_argument("--models", default="",
+                   help="comma filter, e.g. resnet50,bert_large")
     args = p.parse_args()
 
     done = set()
@@ -75,7 +84,10 @@ def main():
     if os.path.exists(args.out):
         reports = json.load(open(args.out))
 
+    want = {m for m in args.models.split(",") if m}
     for driver, model, bs in TASKS:
+        if want and model not in want:
+            continue
         for method in METHOD_SETS[args.set]:
             key = f"{driver}/{model}/bs{bs}/{method}/g{args.gpus}"
             if key in done:
