@@ -56,11 +56,18 @@ class DearOptimizer(torch.optim.Optimizer):
                  num_groups: int = 0,
                  exclude_parts: str = "",
                  backend: Optional[CommBackend] = None,
-                 comm_dtype: Optional[torch.dtype] = None):
+                 comm_dtype: Optional[torch.dtype] = None,
+                 accum_steps: int = 1):
         self.optim = optimizer
         self.model = model
         self.threshold_bytes = threshold_bytes
         self.num_groups = num_groups
+        # gradient accumulation: fire the reduce-scatter only on the
+        # accum_steps-th backward of each group; gradients sum in the bucket
+        # across micro-batches (call step() once per cycle; scale the loss by
+        # 1/accum_steps for averaging, as with DDP).  Capability beyond the
+        # reference, which supports exactly one backward per step.
+        self.accum_steps = max(int(accum_steps), 1)
         # ablation switches, reference dopt_rsag.py:71-72 / batch.sh
         parts = {p.strip() for p in exclude_parts.split(",") if p.strip()}
         self._do_rs = "reducescatter" not in parts
@@ -100,6 +107,7 @@ class DearOptimizer(torch.optim.Optimizer):
         # per-group scheduling state
         n = len(self.groups)
         self._ready_count = [0] * n
+        self._accum_count = [0] * n
         self._rs_handle = [NULL_HANDLE] * n
         self._ag_handle = [NULL_HANDLE] * n
         self._updated = [True] * n   # True => no pending gathered grads to apply
@@ -155,9 +163,14 @@ class DearOptimizer(torch.optim.Optimizer):
                 self._grad_view_fixups += 1
             self._ready_count[group.index] += 1
             if self._ready_count[group.index] == len(group.slots):
-                if self._tracer:
-                    self._tracer.instant(f"rs_launch/g{group.index}", "comm")
-                self._launch_rs(group)
+                self._ready_count[group.index] = 0
+                self._accum_count[group.index] += 1
+                if self._accum_count[group.index] >= self.accum_steps:
+                    self._accum_count[group.index] = 0
+                    if self._tracer:
+                        self._tracer.instant(f"rs_launch/g{group.index}",
+                                             "comm")
+                    self._launch_rs(group)
         return hook
 
     def _launch_rs(self, group: BucketGroup):
@@ -233,6 +246,7 @@ class DearOptimizer(torch.optim.Optimizer):
             self._updated[g.index] = False
         self._prev_iter_done = last
         self._ready_count = [0] * len(self.groups)
+        self._accum_count = [0] * len(self.groups)
         self._num_steps += 1
         return loss
 

@@ -141,3 +141,32 @@ def test_state_dict_roundtrip():
     sd = opt.state_dict()
     assert sd["state"], "optimizer state should be populated"
     opt.load_state_dict(sd)
+
+
+def test_gradient_accumulation_matches_serial_big_batch():
+    """accum_steps=2 with half batches == one step on the full batch (loss
+    scaled by 1/accum so gradients average)."""
+    T, bs = 4, 8
+    data = _data(T=T, bs=2 * bs, seed=9)
+    # serial: full batch per step
+    a = _model()
+    oa = torch.optim.SGD(a.parameters(), lr=0.05, momentum=0.9)
+    for x, y in data:
+        oa.zero_grad()
+        nn.functional.mse_loss(a(x), y).backward()
+        oa.step()
+    # DeAR: two half-batch backwards per step, accum_steps=2
+    b = _model()
+    ob = dear.DistributedOptimizer(
+        torch.optim.SGD(b.parameters(), lr=0.05, momentum=0.9), model=b,
+        threshold_bytes=1 << 12, accum_steps=2)
+    for x, y in data:
+        ob.zero_grad()
+        for k in range(2):
+            xs, ys = x[k * bs:(k + 1) * bs], y[k * bs:(k + 1) * bs]
+            (nn.functional.mse_loss(b(xs), ys) / 2).backward()
+        ob.step()
+    ob.synchronize()
+    for (na, pa), (_, pb) in zip(a.named_parameters(), b.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-5), \
+            f"{na}: {(pa - pb).abs().max():.3e}"
