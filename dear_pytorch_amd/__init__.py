@@ -75,6 +75,13 @@ def local_rank() -> int:
     return int(os.environ.get("LOCAL_RANK", "0"))
 
 
+def barrier():
+    """Host barrier across the world (reference comm_core g_barriar /
+    Communicator.barrier)."""
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.barrier()
+
+
 def shutdown():
     global _generic_backend
     _generic_backend = None

@@ -57,11 +57,13 @@ class DearOptimizer(torch.optim.Optimizer):
                  exclude_parts: str = "",
                  backend: Optional[CommBackend] = None,
                  comm_dtype: Optional[torch.dtype] = None,
-                 accum_steps: int = 1):
+                 accum_steps: int = 1,
+                 num_nearby_layers: Optional[int] = None):
         self.optim = optimizer
         self.model = model
         self.threshold_bytes = threshold_bytes
         self.num_groups = num_groups
+        self.num_nearby_layers = num_nearby_layers
         # gradient accumulation: fire the reduce-scatter only on the
         # accum_steps-th backward of each group; gradients sum in the bucket
         # across micro-batches (call step() once per cycle; scale the loss by
@@ -95,7 +97,8 @@ class DearOptimizer(torch.optim.Optimizer):
     def _build(self, threshold_bytes, fusion_flags=None):
         self.groups: List[BucketGroup] = build_groups(
             self.model, threshold_bytes, self.num_groups,
-            fusion_flags=fusion_flags)
+            fusion_flags=fusion_flags,
+            nearby_layers=self.num_nearby_layers)
         for g in self.groups:
             g.allocate(self.size, self._device,
                        comm_dtype=self.comm_dtype if self.size > 1 else None)

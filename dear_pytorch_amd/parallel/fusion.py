@@ -144,7 +144,8 @@ def _module_param_order(model: nn.Module) -> List[Tuple[nn.Module, List[Tuple[st
 def build_groups(model: nn.Module,
                  threshold_bytes: Optional[int] = 25 * 1024 * 1024,
                  num_groups: int = 0,
-                 fusion_flags: Optional[List[bool]] = None) -> List[BucketGroup]:
+                 fusion_flags: Optional[List[bool]] = None,
+                 nearby_layers: Optional[int] = None) -> List[BucketGroup]:
     """Partition the model's modules (forward order) into bucket groups.
 
     threshold_bytes: close a group once it holds >= threshold bytes of grads
@@ -152,6 +153,9 @@ def build_groups(model: nn.Module,
         group (the no-tensor-fusion ablation).
     num_groups: if > 0, override threshold and split into ~equal-size groups
         (reference num_groups arg, dopt_rsag.py:105-117).
+    nearby_layers: with threshold_bytes=None, merge fixed runs of N
+        consecutive modules per group (reference NUM_NEARBY_LAYERS,
+        dopt_rsag.py:39; N=1 == per-module, the no-fusion ablation).
     fusion_flags: explicit per-module "start new group" booleans (wait-time
         adaptive regrouping, dopt_rsag_wt.py) — length = #modules-with-params.
     """
@@ -164,19 +168,25 @@ def build_groups(model: nn.Module,
     groups: List[BucketGroup] = []
     cur: Optional[BucketGroup] = None
     cur_bytes = 0
+    cur_mods = 0
     for i, (m, ps) in enumerate(mods):
         start_new = cur is None
         if fusion_flags is not None:
             start_new = start_new or fusion_flags[i]
         elif threshold_bytes is None:
-            start_new = True
+            if nearby_layers and nearby_layers > 1:
+                start_new = start_new or cur_mods >= nearby_layers
+            else:
+                start_new = True
         elif cur_bytes >= threshold_bytes:
             start_new = True
         if start_new:
             cur = BucketGroup(index=len(groups), modules=[], slots=[])
             groups.append(cur)
             cur_bytes = 0
+            cur_mods = 0
         cur.modules.append(m)
+        cur_mods += 1
         for name, p in ps:
             off = _align(cur.numel)
             cur.slots.append(ParamSlot(name, p, off, p.numel()))

@@ -110,3 +110,10 @@ def test_channels_last_grad_views_accumulate_in_place():
         o2.step()
     for (na, pa), (_, pb) in zip(m2.named_parameters(), m.named_parameters()):
         assert torch.allclose(pa, pb, atol=1e-6), na
+
+
+def test_nearby_layers_grouping():
+    m = _mlp()
+    groups = build_groups(m, threshold_bytes=None, nearby_layers=2)
+    assert len(groups) == 2  # 3 param modules in runs of 2
+    assert len(groups[0].modules) == 2 and len(groups[1].modules) == 1
