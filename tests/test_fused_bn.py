@@ -164,3 +164,18 @@ def test_gpu_fused_resnet50_e2e_matches_plain():
         assert abs(la.item() - lb.item()) < 0.15 * max(1.0, la.item()), \
             (i, la.item(), lb.item())
     oa.synchronize(); ob.synchronize()
+
+
+def test_fused_densenet_cpu_matches_plain():
+    from dear_pytorch_amd import models
+    torch.manual_seed(0)
+    a = models.get_cnn("densenet121", num_classes=10, fused_bn=False)
+    torch.manual_seed(0)
+    b = models.get_cnn("densenet121", num_classes=10, fused_bn=True)
+    b.load_state_dict(a.state_dict())
+    x = torch.randn(2, 3, 64, 64)
+    ya, yb = a(x), b(x)
+    assert torch.allclose(ya, yb, atol=1e-5)
+    ya.sum().backward(); yb.sum().backward()
+    for (n, pa), (_, pb) in zip(a.named_parameters(), b.named_parameters()):
+        assert torch.allclose(pa.grad, pb.grad, atol=1e-4), n
