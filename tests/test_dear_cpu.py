@@ -170,3 +170,20 @@ def test_gradient_accumulation_matches_serial_big_batch():
     for (na, pa), (_, pb) in zip(a.named_parameters(), b.named_parameters()):
         assert torch.allclose(pa, pb, atol=1e-5), \
             f"{na}: {(pa - pb).abs().max():.3e}"
+
+
+def test_dear_with_autocast_cpu():
+    """--amp path: autocast compute, fp32 grads into buckets."""
+    data = _data(T=3)
+    m = _model()
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05), model=m)
+    for x, y in data:
+        opt.zero_grad()
+        with torch.autocast("cpu", dtype=torch.bfloat16):
+            loss = nn.functional.mse_loss(m(x), y)
+        loss.backward()
+        opt.step()
+    opt.synchronize()
+    assert all(torch.isfinite(p).all() for p in m.parameters())
+    assert all(p.grad.dtype == torch.float32 for p in m.parameters())
