@@ -43,9 +43,11 @@ class WaitTimeAdaptiveFusion:
 
             def hook(*a):
                 outer._push_t[id(p)] = time.perf_counter()
-                inner(*a)
                 g, s = opt._slot_of[p]
-                if opt._ready_count[g.index] == len(g.slots):
+                pre = opt._ready_count[g.index]
+                inner(*a)
+                # the inner hook resets ready_count to 0 on completion
+                if pre == len(g.slots) - 1:
                     done = time.perf_counter()
                     for slot in g.slots:
                         t0 = outer._push_t.get(id(slot.param))
