@@ -300,7 +300,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("stream_handle", &Communicator::stream_handle)
       .def("wait_stream", &Communicator::wait_stream)
       .def("wait_op_stream", &Communicator::wait_op_stream)
-      .def("wait_op_host", &Communicator::wait_op_host)
+      .def("wait_op_host", &Communicator::wait_op_host,
+           py::call_guard<py::gil_scoped_release>())
       .def("op_done", &Communicator::op_done)
       .def("synchronize", &Communicator::synchronize,
            py::call_guard<py::gil_scoped_release>())
