@@ -4,7 +4,11 @@ missing `tcmm.f_topk`, wfbp/dopt.py:95).
 Strategy (no full sort): binary-search an |x| threshold with the count_ge
 kernel (few passes over the data, each HBM-bandwidth-bound), then compact
 matches with select_ge.  Exact k is enforced by trimming the final
-overshoot with a small torch.topk on <= 2k candidates.
+overshoot with a small torch.topk on <= ~2k candidates.
+
+The search/trim logic is backend-injected (`_topk_abs_impl`) so it is unit-
+tested on CPU against a torch reference; `topk_abs_native` binds the real
+CDNA4 kernels.
 """
 from __future__ import annotations
 
@@ -13,22 +17,39 @@ from typing import Tuple
 import torch
 
 
-def topk_abs_native(x: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor]:
-    assert x.is_cuda and x.dtype == torch.float32
-    import dear_pytorch_amd._kernels as K
+class _TorchKernels:
+    """CPU/torch stand-in with the same interface as _kernels (tests)."""
+
+    @staticmethod
+    def count_ge(x, thr, counts):
+        for i, t in enumerate(thr.tolist()):
+            counts[i] += int((x.abs() >= t).sum())
+
+    @staticmethod
+    def select_ge(x, thr, out_idx, out_val, cursor):
+        idx = (x.abs() >= thr).nonzero(as_tuple=False).view(-1)
+        n = min(idx.numel(), out_idx.numel())
+        out_idx[:n] = idx[:n]
+        out_val[:n] = x[idx[:n]]
+        cursor[0] = idx.numel()
+
+
+def _topk_abs_impl(x: torch.Tensor, k: int, K) -> Tuple[torch.Tensor,
+                                                        torch.Tensor]:
     n = x.numel()
     k = max(1, min(k, n))
     if k == n:
         idx = torch.arange(n, device=x.device)
         return x.clone(), idx
-    lo_t, hi_t = 0.0, float(x.abs().max().item())
+    hi_t = float(x.abs().max().item())
     if hi_t == 0.0:
         idx = torch.arange(k, device=x.device)
         return x[:k].clone(), idx
-    # binary search threshold so that count in [k, 2k]
+    # binary search a threshold with count in [k, 2k]
+    lo, hi = 0.0, hi_t
     thr = hi_t / 2
-    lo, hi = lo_t, hi_t
     counts = torch.zeros(1, dtype=torch.int32, device=x.device)
+    c = n
     for _ in range(24):
         counts.zero_()
         K.count_ge(x, torch.tensor([thr], device=x.device), counts)
@@ -40,8 +61,6 @@ def topk_abs_native(x: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor
         else:
             lo = thr
         thr = (lo + hi) / 2
-    else:
-        c = int(counts.item())
     cap = max(c, k) + 64
     out_idx = torch.zeros(cap, dtype=torch.int64, device=x.device)
     out_val = torch.zeros(cap, dtype=torch.float32, device=x.device)
@@ -52,9 +71,16 @@ def topk_abs_native(x: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor
     if m == k:
         return out_val, out_idx
     if m < k:
-        # threshold too high (ties/fp): fall back to torch.topk on the rest
+        # threshold too high (fp ties): exact fallback
         _, idx = torch.topk(x.abs(), k, sorted=False)
         return x[idx], idx
     # trim overshoot: top-k among the m candidates (m <= ~2k, cheap)
     _, sub = torch.topk(out_val.abs(), k, sorted=False)
     return out_val[sub], out_idx[sub]
+
+
+def topk_abs_native(x: torch.Tensor, k: int) -> Tuple[torch.Tensor,
+                                                      torch.Tensor]:
+    assert x.is_cuda and x.dtype == torch.float32
+    import dear_pytorch_amd._kernels as K
+    return _topk_abs_impl(x, k, K)

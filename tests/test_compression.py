@@ -89,3 +89,25 @@ def test_gtopk_ws2():
     (d0, n0), (d1, n1) = outs
     assert torch.equal(d0, d1)
     assert 1 <= n0 <= 8 and n0 == n1
+
+
+def test_topk_impl_logic_cpu_stub():
+    """The native top-k's search/trim logic vs torch.topk, kernels stubbed."""
+    from dear_pytorch_amd.ops.topk import _topk_abs_impl, _TorchKernels
+    g = torch.Generator().manual_seed(4)
+    for n, k in [(1000, 10), (1000, 1), (64, 64), (64, 63), (5000, 500)]:
+        x = torch.randn(n, generator=g)
+        vals, idx = _topk_abs_impl(x, k, _TorchKernels)
+        assert vals.numel() == k and idx.numel() == k
+        assert torch.equal(x[idx], vals)
+        ref_vals, _ = torch.topk(x.abs(), k, sorted=True)
+        got = vals.abs().sort(descending=True).values
+        assert torch.allclose(got, ref_vals), (n, k)
+    # all-zero input
+    z = torch.zeros(100)
+    vals, idx = _topk_abs_impl(z, 5, _TorchKernels)
+    assert vals.numel() == 5
+    # duplicates / ties
+    t = torch.ones(50)
+    vals, idx = _topk_abs_impl(t, 7, _TorchKernels)
+    assert vals.numel() == 7 and (vals == 1).all()
