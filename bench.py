@@ -19,6 +19,11 @@ import time
 # are wanted).
 os.environ.setdefault("MIOPEN_FIND_MODE", os.environ.get("DEAR_MIOPEN_FIND",
                                                          "FAST"))
+# a committed tuned perf-DB (tools/miopen_tune.sh) is picked up automatically
+_udb = os.path.join(os.path.dirname(os.path.abspath(__file__)), "miopen_udb")
+if os.path.isdir(_udb):
+    os.environ.setdefault("MIOPEN_USER_DB_PATH", _udb)
+    os.environ.setdefault("MIOPEN_CUSTOM_CACHE_DIR", _udb)
 
 import torch  # noqa: E402
 
