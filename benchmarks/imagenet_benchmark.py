@@ -42,6 +42,9 @@ def main():
                    help="bf16 autocast compute (reference --fp16 axis; "
                         "gradients/optimizer stay fp32)")
     args = p.parse_args()
+    if args.timeline:
+        os.environ["DEAR_TIMELINE"] = args.timeline  # chrome-trace of the
+        # DeAR queue events (reference WFSGD_TIMELINE via horovod_mpi_cj.sh)
 
     import dear_pytorch_amd as dear
     from dear_pytorch_amd import models
@@ -152,6 +155,12 @@ def main():
     log(f"Img/sec per GPU: {img_sec_mean:.1f} +-{img_sec_conf:.1f}")
     log(f"Total img/sec on {world} GPU(s): {world * img_sec_mean:.1f} "
         f"+-{world * img_sec_conf:.1f}")
+    if args.timeline:
+        from dear_pytorch_amd.profiling import tracer
+        t = tracer()
+        if t:
+            t.save()
+            log(f"timeline written to {t.path}")
     dear.shutdown()
 
 

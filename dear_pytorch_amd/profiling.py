@@ -176,6 +176,24 @@ class ChromeTracer:
         self.save()
 
 
+def torch_profile(step_fn, path: str, steps: int = 3, warmup: int = 2):
+    """Profile `steps` calls of step_fn with torch.profiler (CPU + HIP
+    activities on ROCm) and export a chrome trace to `path` (the MI355X
+    equivalent of the reference's nvprof scripts, horovod/prof.sh)."""
+    from torch.profiler import profile, schedule, ProfilerActivity
+    acts = [ProfilerActivity.CPU]
+    if torch.cuda.is_available():
+        acts.append(ProfilerActivity.CUDA)
+    with profile(activities=acts,
+                 schedule=schedule(wait=0, warmup=warmup, active=steps),
+                 record_shapes=False) as prof:
+        for _ in range(warmup + steps):
+            step_fn()
+            prof.step()
+    prof.export_chrome_trace(path)
+    return prof
+
+
 _tracer: Optional[ChromeTracer] = None
 
 

@@ -150,3 +150,36 @@ def test_checkpoint_save_load_roundtrip(tmp_path):
     for (ka, va), (_, vb) in zip(m.state_dict().items(),
                                  m2.state_dict().items()):
         assert torch.allclose(va, vb, atol=1e-6), ka
+
+
+def test_torch_profile_wrapper(tmp_path):
+    from dear_pytorch_amd.profiling import torch_profile
+    m = _model()
+    x, y = torch.randn(4, 32), torch.randn(4, 8)
+
+    def step():
+        m.zero_grad()
+        nn.functional.mse_loss(m(x), y).backward()
+
+    p = tmp_path / "tp.json"
+    torch_profile(step, str(p), steps=2, warmup=1)
+    import json
+    data = json.load(open(p))
+    assert "traceEvents" in data and len(data["traceEvents"]) > 10
+
+
+def test_driver_timeline_flag(tmp_path):
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = tmp_path / "tl.json"
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "benchmarks",
+                                      "imagenet_benchmark.py"),
+         "--model", "resnet18", "--batch-size", "2",
+         "--num-warmup-batches", "1", "--num-batches-per-iter", "1",
+         "--num-iters", "1", "--timeline", str(out)],
+        capture_output=True, text=True, timeout=240, cwd=repo)
+    assert r.returncode == 0, r.stderr[-1200:]
+    import json
+    ev = json.load(open(out))["traceEvents"]
+    assert any("rs_launch" in e.get("name", "") for e in ev)
