@@ -154,6 +154,10 @@ def main():
         torch.cuda.set_device(device)
         torch.backends.cudnn.benchmark = True
 
+    if args.gpus != world and rank == 0:
+        import sys
+        print(f"[bench] note: --gpus {args.gpus} but WORLD_SIZE={world}; "
+              f"using the launched world", file=sys.stderr, flush=True)
     model, opt_fn, step_fn, bs, unit, metric = build_workload(args, device)
     if world > 1:
         dear.broadcast_parameters(model.state_dict(), root_rank=0)
