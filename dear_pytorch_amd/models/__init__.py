@@ -29,7 +29,7 @@ def get_cnn(name: str, num_classes: int = 1000, fused_bn: bool = False):
         raise KeyError(f"unknown model '{name}'; have {sorted(_REGISTRY)}")
     if name == "mnistnet":
         return _REGISTRY[name]()
-    if name.startswith(("resnet", "densenet")):
+    if name.startswith(("resnet", "densenet", "inception")):
         return _REGISTRY[name](num_classes=num_classes, fused_bn=fused_bn)
     return _REGISTRY[name](num_classes=num_classes)
 

@@ -7,12 +7,20 @@ import torch.nn as nn
 __all__ = ["inceptionv4"]
 
 
+_FUSED_BN = False  # set by inceptionv4(fused_bn=...) during construction
+
+
 class ConvBN(nn.Module):
     def __init__(self, cin, cout, k, stride=1, padding=0):
         super().__init__()
         self.conv = nn.Conv2d(cin, cout, k, stride, padding, bias=False)
-        self.bn = nn.BatchNorm2d(cout, eps=1e-3)
-        self.act = nn.ReLU(inplace=True)
+        if _FUSED_BN and cout % 4 == 0:
+            from ..ops.fused_bn import FusedBNAct2d
+            self.bn = FusedBNAct2d(cout, relu=True, eps=1e-3)
+            self.act = nn.Identity()
+        else:
+            self.bn = nn.BatchNorm2d(cout, eps=1e-3)
+            self.act = nn.ReLU(inplace=True)
 
     def forward(self, x):
         return self.act(self.bn(self.conv(x)))
@@ -143,5 +151,10 @@ class InceptionV4(nn.Module):
         return self.classifier(self.dropout(x))
 
 
-def inceptionv4(num_classes=1000):
-    return InceptionV4(num_classes)
+def inceptionv4(num_classes=1000, fused_bn=False):
+    global _FUSED_BN
+    _FUSED_BN = fused_bn
+    try:
+        return InceptionV4(num_classes)
+    finally:
+        _FUSED_BN = False

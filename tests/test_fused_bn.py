@@ -179,3 +179,17 @@ def test_fused_densenet_cpu_matches_plain():
     ya.sum().backward(); yb.sum().backward()
     for (n, pa), (_, pb) in zip(a.named_parameters(), b.named_parameters()):
         assert torch.allclose(pa.grad, pb.grad, atol=1e-4), n
+
+
+def test_fused_inceptionv4_cpu_matches_plain():
+    from dear_pytorch_amd import models
+    torch.manual_seed(0)
+    a = models.get_cnn("inceptionv4", num_classes=10, fused_bn=False)
+    torch.manual_seed(0)
+    b = models.get_cnn("inceptionv4", num_classes=10, fused_bn=True)
+    b.load_state_dict(a.state_dict())
+    a.eval(); b.eval()  # cheap single pass, eval avoids dropout
+    x = torch.randn(1, 3, 299, 299)
+    with torch.no_grad():
+        ya, yb = a(x), b(x)
+    assert torch.allclose(ya, yb, atol=1e-5)
