@@ -296,6 +296,21 @@ class DearOptimizer(torch.optim.Optimizer):
     def load_state_dict(self, sd):
         self.optim.load_state_dict(sd)
 
+    def summary(self) -> str:
+        """Human-readable fusion-plan table (group sizes, module spans, wire
+        format) — the reference printed this at init (dopt_rsag.py:175-180)."""
+        lines = [f"DeAR plan: {len(self.groups)} groups, world={self.size}, "
+                 f"threshold={self.threshold_bytes}, "
+                 f"wire={self.comm_dtype or 'fp32'}, "
+                 f"accum={self.accum_steps}"]
+        for g in self.groups:
+            mods = type(g.modules[0]).__name__
+            if len(g.modules) > 1:
+                mods += f"..{type(g.modules[-1]).__name__}"
+            lines.append(f"  g{g.index}: {len(g.slots):3d} tensors "
+                         f"{g.nbytes / 1e6:8.1f} MB  [{mods}]")
+        return "\n".join(lines)
+
     def __repr__(self):
         return (f"DearOptimizer(groups={len(self.groups)}, size={self.size}, "
                 f"inner={type(self.optim).__name__})")

@@ -384,3 +384,37 @@ def test_regroup_midtraining_ws2_matches_serial():
         assert torch.allclose(ref[k], outs[0][k], atol=1e-5), \
             f"{k}: {(ref[k] - outs[0][k]).abs().max():.3e}"
         assert torch.equal(outs[0][k], outs[1][k])
+
+
+def _rank_checkpoint(rank, world, path):
+    import torch
+    import dear_pytorch_amd as dear
+    from dear_pytorch_amd import checkpoint
+    dear.init(backend="gloo")
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m)
+    for x, y in _full_data(3, 8):
+        xs, ys = x[rank * 8:(rank + 1) * 8], y[rank * 8:(rank + 1) * 8]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    checkpoint.save(path, m, opt, step=3)
+    # resume into fresh replicas (deliberately divergent inits)
+    m2 = _model(seed=17 + rank)
+    opt2 = dear.DistributedOptimizer(
+        torch.optim.SGD(m2.parameters(), lr=0.05, momentum=0.9), model=m2)
+    step = checkpoint.load(path, m2, opt2.optim)
+    assert step == 3
+    out = {k: v.clone() for k, v in m2.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_checkpoint_save_resume_ws2(tmp_path):
+    path = str(tmp_path / "ck.pt")
+    outs = run_dist(_rank_checkpoint, world_size=2, args=(path,))
+    for k in outs[0]:
+        assert torch.equal(outs[0][k], outs[1][k]), k
