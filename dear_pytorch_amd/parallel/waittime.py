@@ -101,7 +101,7 @@ class WaitTimeAdaptiveFusion:
     def _sync_flags(self, flags: List[bool]) -> List[bool]:
         import torch.distributed as dist
         if dist.is_initialized() and dist.get_world_size() > 1:
-            t = torch.tensor([1.0 if f else 0.0 for f in flags])
-            dist.broadcast(t, src=0)
-            return [bool(v) for v in t.tolist()]
+            from ..utils.dist_helpers import bcast_floats
+            vals = bcast_floats([1.0 if f else 0.0 for f in flags])
+            return [bool(v) for v in vals]
         return flags

@@ -125,10 +125,8 @@ class ThresholdTuner:
     def _sync_threshold(self, mb: float) -> float:
         """Rank-0's threshold wins (reference bcasts via MPI,
         dopt_rsag_bo.py:153)."""
-        import torch
         import torch.distributed as dist
         if dist.is_initialized() and dist.get_world_size() > 1:
-            t = torch.tensor([mb], dtype=torch.float64)
-            dist.broadcast(t, src=0)
-            return float(t.item())
+            from .utils.dist_helpers import bcast_floats
+            return float(bcast_floats([mb])[0])
         return mb

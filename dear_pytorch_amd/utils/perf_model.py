@@ -102,10 +102,9 @@ def plan_mgwfbp_flags(model: torch.nn.Module, backend=None,
             acc_bytes = 0
             remaining_tau = 0.0
     if backend is not None and world > 1:
-        import torch.distributed as dist
-        t = torch.tensor([1.0 if f else 0.0 for f in flags])
-        dist.broadcast(t, src=0)
-        flags = [bool(v) for v in t.tolist()]
+        from .dist_helpers import bcast_floats
+        flags = [bool(v) for v in bcast_floats([1.0 if f else 0.0
+                                                for f in flags])]
     return flags
 
 
