@@ -15,6 +15,12 @@ __all__ = ["ReduceBcastOptimizer"]
 class ReduceBcastOptimizer(DearOptimizer):
     ROOT = 0
 
+    def __init__(self, *a, **kw):
+        if kw.get("comm_dtype") is not None:
+            raise ValueError("reduce+broadcast ablation supports fp32 wire "
+                             "format only")
+        super().__init__(*a, **kw)
+
     def _launch_rs(self, group: BucketGroup):
         if self.size > 1 and self._do_rs:
             self._rs_handle[group.index] = self.comm_rs.reduce(
