@@ -418,3 +418,46 @@ def test_checkpoint_save_resume_ws2(tmp_path):
     outs = run_dist(_rank_checkpoint, world_size=2, args=(path,))
     for k in outs[0]:
         assert torch.equal(outs[0][k], outs[1][k]), k
+
+
+def _rank_cnn_bn(rank, world):
+    import torch
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    torch.manual_seed(0)
+
+    def make():
+        torch.manual_seed(0)
+        return nn.Sequential(
+            nn.Conv2d(3, 8, 3, padding=1), nn.BatchNorm2d(8), nn.ReLU(),
+            nn.Conv2d(8, 8, 3, padding=1), nn.BatchNorm2d(8), nn.ReLU(),
+            nn.AdaptiveAvgPool2d(1), nn.Flatten(), nn.Linear(8, 4))
+    m = make()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m,
+        threshold_bytes=256)
+    g = torch.Generator().manual_seed(3)
+    data = [(torch.randn(2 * 4, 3, 8, 8, generator=g),
+             torch.randn(2 * 4, 4, generator=g)) for _ in range(4)]
+    for x, y in data:
+        xs, ys = x[rank * 4:(rank + 1) * 4], y[rank * 4:(rank + 1) * 4]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    opt.synchronize()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_cnn_batchnorm_dear_ws2_ranks_agree():
+    """Conv+BN model under DeAR ws2: parameters identical across ranks;
+    BN running stats differ per rank (local batches) as with DDP defaults."""
+    outs = run_dist(_rank_cnn_bn, world_size=2)
+    for k in outs[0]:
+        if "running" in k or "num_batches" in k:
+            continue
+        assert torch.equal(outs[0][k], outs[1][k]), k
+        assert torch.isfinite(outs[0][k]).all(), k
