@@ -11,8 +11,7 @@ error feedback.
 """
 from __future__ import annotations
 
-import math
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 

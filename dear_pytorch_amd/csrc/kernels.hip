@@ -31,13 +31,8 @@ namespace {
 
 constexpr int kBlock = 256;
 
-// Chunk descriptor row: {param_ptr, bucket_off, n} as int64 (ops/fused.py).
-struct __align__(16) Chunk {
-  const int64_t ptr;
-  const int64_t off;
-  const int64_t n;
-};
-
+// Chunk descriptor rows: {param_ptr, bucket_off, n} int64 triples
+// (built by ops/fused._build_desc, read directly from the desc array).
 __device__ __forceinline__ bool aligned16(const void* p) {
   return (reinterpret_cast<uintptr_t>(p) & 15u) == 0;
 }

@@ -6,7 +6,6 @@ from bert_config.json + BertPretrainingCriterion).  Configs: Base 12L/768h/12
 heads, Large 24L/1024h/16 heads, vocab padded to a multiple of 8 (30528).
 Attention runs through torch.nn.functional.scaled_dot_product_attention
 (MIOpen/CK fused path on ROCm)."""
-import math
 from dataclasses import dataclass
 
 import torch

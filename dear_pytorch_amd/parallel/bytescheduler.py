@@ -24,7 +24,7 @@ from typing import List, Optional
 
 import torch
 
-from ..comm.backend import CommBackend, create_backend, NULL_HANDLE
+from ..comm.backend import CommBackend, create_backend
 from .fusion import BucketGroup, build_groups
 
 __all__ = ["ByteSchedulerOptimizer"]

@@ -5,8 +5,6 @@ from __future__ import annotations
 
 from typing import List
 
-import torch.nn as nn
-
 from .dear import DearOptimizer
 from .fusion import BucketGroup, ParamSlot, _module_param_order
 
