@@ -324,7 +324,13 @@ def DistributedOptimizer(optimizer: torch.optim.Optimizer,
                          num_groups: int = 0,
                          exclude_parts: str = "",
                          **kw) -> DearOptimizer:
-    """Horovod-shaped factory (reference dopt_rsag.py:377-394)."""
+    """Horovod-shaped factory (reference dopt_rsag.py:377-394).
+
+    ``named_parameters`` and ``compression`` are accepted for signature parity
+    with the reference/Horovod surface; fusion groups are planned from
+    ``model`` directly and gradient compression lives in the WFBP path
+    (compression.py), so both are ignored here.
+    """
     assert model is not None, "DeAR needs the model to plan fusion groups"
     return DearOptimizer(optimizer, model, threshold_bytes=threshold_bytes,
                          num_groups=num_groups, exclude_parts=exclude_parts, **kw)
