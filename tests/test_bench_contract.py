@@ -34,3 +34,18 @@ def test_bench_json_contract_cpu():
     cfg = d["config"]
     assert cfg["model"] == "resnet18" and cfg["parallelism"] == "dp1"
     assert cfg["global_batch"] == 2
+
+
+def test_matrix_runner_contract_parsing():
+    sys.path.insert(0, os.path.join(REPO, "benchmarks"))
+    import run_matrix
+    txt = "noise\nImg/sec per GPU: 10 +-1\nTotal img/sec on 8 GPU(s): 12345.6 +-7.8\n"
+    assert run_matrix.extract_total(txt) == 12345.6
+    assert run_matrix.extract_total("nothing here") is None
+    cmd = run_matrix.gen_cmd("imagenet", "resnet50", 64, "dear-notf", 8, 5)
+    assert "--no-fusion" in cmd and "--method" in cmd
+    cmd = run_matrix.gen_cmd("bert", "bert_large", 32, "mgwfbp", 4, 5)
+    assert "bert_benchmark.py" in " ".join(cmd)
+    assert "--nproc-per-node=4" in " ".join(cmd)
+    cmd = run_matrix.gen_cmd("imagenet", "resnet50", 64, "dear-nors", 8, 5)
+    assert "reducescatter" in cmd
