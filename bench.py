@@ -39,7 +39,8 @@ def parse_args():
                    help="per-GPU batch (default: 64 CNN / 32 BERT)")
     p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--method", default="dear",
-                   choices=["dear", "ddp", "wfbp", "mgwfbp", "naive", "rb", "bytescheduler"],
+                   choices=["dear", "dear-bo", "ddp", "wfbp", "mgwfbp",
+                            "naive", "rb", "bytescheduler"],
                    help="gradient-sync method (dear is the product)")
     p.add_argument("--threshold-mb", type=float, default=25.0)
     p.add_argument("--exclude-parts", default="")
@@ -128,13 +129,35 @@ def wrap_method(args, model, opt_fn):
                 gradient_as_bucket_view=True)
         opt = opt_fn(model.parameters())
         return model, opt
-    if args.method == "dear":
+    if args.method in ("dear", "dear-bo"):
         cdt = {"fp32": None, "bf16": torch.bfloat16,
                "fp16": torch.float16}[args.comm_dtype]
         opt = dear.DistributedOptimizer(opt_fn(model.parameters()),
                                         model=model, threshold_bytes=threshold,
                                         exclude_parts=args.exclude_parts,
                                         comm_dtype=cdt)
+        if args.method == "dear-bo":
+            from dear_pytorch_amd.tuner import ThresholdTuner
+            tuner = ThresholdTuner(opt, window=3,
+                                   warmup=min(args.warmup, 5), trials=8,
+                                   verbose=False)
+            inner = opt
+
+            class _TunedStep:
+                """step() facade driving the BO tuner around each iteration."""
+
+                def __getattr__(self, k):
+                    return getattr(inner, k)
+
+                def zero_grad(self, *a, **k):
+                    tuner.step_begin()
+                    inner.zero_grad()
+
+                def step(self, *a, **k):
+                    r = inner.step()
+                    tuner.step_end()
+                    return r
+            return model, _TunedStep()
         return model, opt
     from dear_pytorch_amd.parallel import baselines
     opt = baselines.make(args.method, opt_fn(model.parameters()), model,
