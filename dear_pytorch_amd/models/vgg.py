@@ -35,6 +35,18 @@ class VGG(nn.Module):
             nn.Linear(512 * 7 * 7, 4096), nn.ReLU(True), nn.Dropout(),
             nn.Linear(4096, 4096), nn.ReLU(True), nn.Dropout(),
             nn.Linear(4096, num_classes))
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out",
+                                        nonlinearity="relu")
+                if m.bias is not None:
+                    nn.init.zeros_(m.bias)
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.ones_(m.weight)
+                nn.init.zeros_(m.bias)
+            elif isinstance(m, nn.Linear):
+                nn.init.normal_(m.weight, 0, 0.01)
+                nn.init.zeros_(m.bias)
 
     def forward(self, x):
         x = self.avgpool(self.features(x)).flatten(1)
