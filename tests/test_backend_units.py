@@ -40,3 +40,17 @@ def test_local_send_recv_raises():
         raise AssertionError("expected RuntimeError")
     except RuntimeError:
         pass
+
+
+def test_public_api_single_process():
+    import dear_pytorch_amd as dear
+    dear.init()  # no WORLD_SIZE: single-process mode
+    assert dear.rank() == 0 and dear.size() == 1
+    assert dear.local_rank() == 0
+    assert not dear.is_initialized()  # no PG created in single-process mode
+    dear.barrier()  # no-op
+    t = dear.allreduce(torch.tensor([2.0]), average=True)
+    assert t.item() == 2.0
+    assert dear.broadcast_object({"a": 1}) == {"a": 1}
+    dear.broadcast_parameters({"w": torch.ones(2)})  # no-op at ws 1
+    dear.shutdown()
