@@ -49,3 +49,21 @@ def test_matrix_runner_contract_parsing():
     assert "--nproc-per-node=4" in " ".join(cmd)
     cmd = run_matrix.gen_cmd("imagenet", "resnet50", 64, "dear-nors", 8, 5)
     assert "reducescatter" in cmd
+
+
+@pytest.mark.timeout(600)
+def test_bench_ws2_feature_flags_cpu():
+    """bench CLI over torchrun ws=2 (gloo) with reduced-precision wire and
+    the BO-tuned method — the flag paths a scale run may use."""
+    base = [sys.executable, "-m", "torch.distributed.run", "--standalone",
+            "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+            os.path.join(REPO, "bench.py"), "--model", "resnet18",
+            "--batch-size", "2", "--steps", "2", "--warmup", "1",
+            "--no-channels-last"]
+    for extra in (["--comm-dtype", "fp16"], ["--method", "dear-bo"]):
+        r = subprocess.run(base + extra, capture_output=True, text=True,
+                           timeout=280, cwd=REPO)
+        assert r.returncode == 0, (extra, r.stderr[-1500:])
+        line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+        d = json.loads(line)
+        assert d["n_gpus"] == 2 and d["value"] > 0
