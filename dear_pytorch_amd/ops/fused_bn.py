@@ -95,7 +95,7 @@ class FusedBNAct2d(nn.BatchNorm2d):
     def __init__(self, num_features, relu=False, **kw):
         super().__init__(num_features, **kw)
         self.relu = relu
-        self._ws = None
+        self._ws = {}  # rows -> _Workspace (variable batch sizes alternate)
 
     def _fast_ok(self, x, residual):
         return (x.is_cuda and x.dtype == torch.float32
@@ -112,13 +112,14 @@ class FusedBNAct2d(nn.BatchNorm2d):
             mom = self.momentum if self.momentum is not None else 0.1
             N, C, H, W = x.shape
             rows = N * H * W
-            if self._ws is None or self._ws.rows != rows \
-                    or self._ws.mean.device != x.device:
-                self._ws = _Workspace(rows, C, x.device)
+            ws = self._ws.get(rows)
+            if ws is None or ws.mean.device != x.device:
+                ws = _Workspace(rows, C, x.device)
+                self._ws[rows] = ws
             return _FusedBNFn.apply(x, self.weight, self.bias,
                                     self.running_mean, self.running_var,
                                     self.training, mom, self.eps, self.relu,
-                                    residual, self._ws)
+                                    residual, ws)
         # reference-numerics fallback (CPU, fp16, NCHW, no-affine...)
         y = super().forward(x)
         if residual is not None:
