@@ -59,7 +59,11 @@ class _FusedBNFn(torch.autograd.Function):
         K.bn_fwd(x, residual, y, weight, bias, mean, invstd,
                  running_mean, running_var, ws.psum, ws.psumsq, rows, C,
                  eps, momentum, training, relu)
-        ctx.save_for_backward(x, y, weight, mean, invstd)
+        # save CLONES of the per-channel stats (C floats, negligible): the
+        # workspace tensors are overwritten by the next forward, which would
+        # silently corrupt a backward that runs after it (multi-forward
+        # patterns like gradient accumulation)
+        ctx.save_for_backward(x, y, weight, mean.clone(), invstd.clone())
         ctx.relu = relu
         ctx.has_res = residual is not None
         ctx.dims = (rows, C)
