@@ -461,3 +461,46 @@ def test_cnn_batchnorm_dear_ws2_ranks_agree():
             continue
         assert torch.equal(outs[0][k], outs[1][k]), k
         assert torch.isfinite(outs[0][k]).all(), k
+
+
+def _rank_bo_stress(rank, world):
+    import torch
+    import dear_pytorch_amd as dear
+    from dear_pytorch_amd.tuner import ThresholdTuner
+    import torch.distributed as dist
+    dear.init(backend="gloo")
+    torch.manual_seed(0)
+    m = nn.Sequential(nn.Linear(64, 128), nn.ReLU(), nn.Linear(128, 128),
+                      nn.ReLU(), nn.Linear(128, 8))
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.01, momentum=0.9), model=m,
+        threshold_bytes=1 << 14)
+    tuner = ThresholdTuner(opt, bounds_mb=(0.005, 0.5), window=2, warmup=2,
+                           trials=8, verbose=False)
+    g = torch.Generator().manual_seed(1)
+    x = torch.randn(8, 64, generator=g)
+    y = torch.randn(8, 8, generator=g)
+    for _ in range(60):
+        tuner.step_begin()
+        opt.zero_grad()
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+        tuner.step_end()
+    opt.synchronize()
+    assert tuner.locked
+    for p in m.parameters():
+        c = p.detach().clone()
+        dist.broadcast(c, src=0)
+        assert torch.equal(c, p.detach()), "rank divergence under regrouping"
+    out = all(torch.isfinite(p).all() for p in m.parameters())
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(600)
+def test_bo_tuning_stress_ws2_stays_consistent():
+    """60 iterations with ~8 live regroups (the riskiest path: buffer
+    teardown + hook re-registration mid-training) must keep ranks
+    bit-identical."""
+    assert all(run_dist(_rank_bo_stress, world_size=2))
