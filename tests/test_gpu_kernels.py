@@ -131,6 +131,20 @@ def test_rccl_communicator_single_rank():
     out = torch.empty(1000, device=_dev())
     comm.wait_op_host(comm.all_gather(shard, out))
     assert torch.allclose(out, ref)
+    # decoupled-allreduce equivalence (reference test_comm.py
+    # decoupleallreduce norm check, on the 17-elem odd size): at world 1 the
+    # RB and RSAG decompositions must equal the plain allreduce exactly
+    t17 = torch.randn(17, device=_dev())
+    a = t17.clone()
+    comm.wait_op_host(comm.all_reduce(a))
+    b = t17.clone()
+    comm.wait_op_host(comm.all_reduce_rb(b, 0))
+    assert torch.allclose(a, b), (a - b).norm().item()
+    pad = torch.zeros(64, device=_dev())
+    pad[:17] = t17
+    sh = torch.empty(64, device=_dev())
+    comm.wait_op_host(comm.all_reduce_rsag(pad, sh))
+    assert torch.allclose(pad[:17], a), (pad[:17] - a).norm().item()
     comm.synchronize()
 
 
