@@ -113,8 +113,13 @@ def broadcast_parameters(params, root_rank: int = 0):
     else:
         items = sorted(dict(params).items())
     be = _generic()
-    handles = [be.broadcast(v.data, root_rank) for _, v in items
-               if torch.is_tensor(v)]
+    tensors = [v.data for _, v in items if torch.is_tensor(v)]
+    from .comm.backend import RcclBackend
+    if isinstance(be, RcclBackend):
+        # one grouped RCCL call (ncclGroupStart/End) instead of N round trips
+        be.comm.wait_op_host(be.comm.broadcast_many(tensors, root_rank))
+        return
+    handles = [be.broadcast(t, root_rank) for t in tensors]
     for h in handles:
         h.host_wait()
 
