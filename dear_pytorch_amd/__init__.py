@@ -116,7 +116,9 @@ def broadcast_parameters(params, root_rank: int = 0):
     tensors = [v.data for _, v in items if torch.is_tensor(v)]
     from .comm.backend import RcclBackend
     if isinstance(be, RcclBackend):
-        # one grouped RCCL call (ncclGroupStart/End) instead of N round trips
+        # one grouped RCCL call (ncclGroupStart/End) instead of N round trips;
+        # order behind the compute stream that initialized the params
+        be.record_compute()
         be.comm.wait_op_host(be.comm.broadcast_many(tensors, root_rank))
         return
     handles = [be.broadcast(t, root_rank) for t in tensors]
