@@ -107,7 +107,17 @@ class DearOptimizer(torch.optim.Optimizer):
         for g in self.groups:
             for s in g.slots:
                 self._slot_of[s.param] = (g, s)
-        # per-group scheduling state
+        self._init_sched_state()
+        if self.rank == 0 and os.environ.get("DEAR_QUIET", "0") != "1":
+            import sys
+            mb = [g.nbytes / 1e6 for g in self.groups]
+            print(f"[dear] {n} fusion groups, sizes MB: "
+                  f"{', '.join(f'{m:.1f}' for m in mb)}", file=sys.stderr,
+                  flush=True)
+
+    def _init_sched_state(self):
+        """Per-group scheduling state, shared by every _build implementation
+        (DearOptimizer, NaiveDearOptimizer) so ablations can't drift."""
         n = len(self.groups)
         self._ready_count = [0] * n
         self._accum_count = [0] * n
@@ -115,12 +125,6 @@ class DearOptimizer(torch.optim.Optimizer):
         self._ag_handle = [NULL_HANDLE] * n
         self._updated = [True] * n   # True => no pending gathered grads to apply
         self._prev_iter_done = NULL_HANDLE
-        if self.rank == 0 and os.environ.get("DEAR_QUIET", "0") != "1":
-            import sys
-            mb = [g.nbytes / 1e6 for g in self.groups]
-            print(f"[dear] {n} fusion groups, sizes MB: "
-                  f"{', '.join(f'{m:.1f}' for m in mb)}", file=sys.stderr,
-                  flush=True)
 
     def _register_hooks(self):
         # per-parameter grad-accumulator hooks -> reduce-scatter on

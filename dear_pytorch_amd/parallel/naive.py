@@ -16,8 +16,9 @@ class NaiveDearOptimizer(DearOptimizer):
         kw.pop("threshold_bytes", None)
         super().__init__(optimizer, model, threshold_bytes=None, **kw)
 
-    def _build(self, threshold_bytes):
-        # one group per parameter tensor
+    def _build(self, threshold_bytes, fusion_flags=None):
+        # one group per parameter tensor (fusion_flags are meaningless without
+        # fusion; accepted so regroup()'s call signature matches)
         groups: List[BucketGroup] = []
         for m, ps in _module_param_order(self.model):
             for name, p in ps:
@@ -27,12 +28,7 @@ class NaiveDearOptimizer(DearOptimizer):
                 groups.append(g)
         self.groups = groups
         for g in self.groups:
-            g.allocate(self.size, self._device)
+            g.allocate(self.size, self._device,
+                       comm_dtype=self.comm_dtype if self.size > 1 else None)
         self._slot_of = {g.slots[0].param: (g, g.slots[0]) for g in self.groups}
-        n = len(self.groups)
-        from ..comm.backend import NULL_HANDLE
-        self._ready_count = [0] * n
-        self._rs_handle = [NULL_HANDLE] * n
-        self._ag_handle = [NULL_HANDLE] * n
-        self._updated = [True] * n
-        self._prev_iter_done = NULL_HANDLE
+        self._init_sched_state()
