@@ -165,6 +165,21 @@ def wrap_method(args, model, opt_fn):
     return model, opt
 
 
+def _provenance_str(args):
+    """Which comm backend actually served each channel (honest-numbers
+    contract: 'rccl-native', 'torch-dist' or 'local').
+    DDP runs on torch's own ProcessGroupNCCL."""
+    if args.method == "ddp":
+        return "ddp/torch-nccl"
+    from dear_pytorch_amd.comm.backend import backend_provenance
+    prov = backend_provenance()
+    if not prov:
+        return "none"
+    kinds = sorted(set(prov.values()))
+    return kinds[0] if len(kinds) == 1 else \
+        ",".join(f"{k}={v}" for k, v in sorted(prov.items()))
+
+
 def main():
     args = parse_args()
     import dear_pytorch_amd as dear
@@ -176,6 +191,12 @@ def main():
     if on_gpu:
         torch.cuda.set_device(device)
         torch.backends.cudnn.benchmark = True
+
+    # on GPU the in-house methods must run their native RCCL channels — a
+    # silent fallback to torch-dist would invalidate the measured number
+    # (VERDICT r1); explicit DEAR_STRICT_NATIVE_COMM=0 opts out.
+    if on_gpu and world > 1 and args.method != "ddp":
+        os.environ.setdefault("DEAR_STRICT_NATIVE_COMM", "1")
 
     if args.gpus != world and rank == 0:
         import sys
@@ -253,6 +274,7 @@ def main():
                 "parallelism": f"dp{world}",
                 "method": args.method,
                 "threshold_mb": None if args.no_fusion else args.threshold_mb,
+                "comm_backend": _provenance_str(args),
             },
         }
         print(json.dumps(out), flush=True)

@@ -36,6 +36,7 @@ __all__ = [
     "TorchDistBackend",
     "RcclBackend",
     "create_backend",
+    "backend_provenance",
 ]
 
 
@@ -301,27 +302,66 @@ class RcclBackend(CommBackend):
         self.comm.synchronize()
 
 
+# Which backend kind actually serves each channel tag, for honest benchmark
+# provenance (a silent native→torch fallback must not be able to
+# launder bench numbers).  bench.py embeds this in its JSON line.
+PROVENANCE: dict = {}
+
+
+def backend_provenance() -> dict:
+    """tag -> 'rccl-native' | 'torch-dist' | 'local' for every channel created."""
+    return dict(PROVENANCE)
+
+
+def _native_precheck() -> bool:
+    """Per-rank, non-collective conditions for the native RCCL path."""
+    if not torch.cuda.is_available() or dist.get_backend() == "gloo":
+        return False
+    if os.environ.get("DEAR_FORCE_TORCH_COMM", "0") == "1":
+        return False
+    try:
+        import dear_pytorch_amd._comm_core  # noqa: F401
+    except Exception:  # noqa: BLE001
+        return False
+    return True
+
+
 def create_backend(tag: str = "generic") -> CommBackend:
     """Pick the backend for this process: native RCCL on ROCm GPUs, torch.distributed
-    (gloo) on CPU, LocalBackend when not distributed."""
+    (gloo) on CPU, LocalBackend when not distributed.
+
+    The native/torch decision is COLLECTIVE: every rank all-reduces its
+    precheck result so all take the same branch (a per-rank fallback would
+    leave some ranks in dist.new_group() and others in ncclCommInitRank —
+    desync/hang).  Once all ranks agree to go native, an init
+    failure raises on every rank instead of falling back, because a partial
+    native init cannot be unwound collectively.
+    """
     if not dist.is_initialized() or dist.get_world_size() == 1:
+        PROVENANCE[tag] = "local"
         return LocalBackend()
+    can_native = _native_precheck()
+    flag = torch.tensor([1 if can_native else 0], dtype=torch.int64)
+    if dist.get_backend() != "gloo" and torch.cuda.is_available():
+        flag = flag.to(torch.device("cuda", torch.cuda.current_device()))
+    dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+    all_native = bool(flag.item())
+    if can_native and not all_native:
+        import sys
+        print(f"[dear] WARNING: channel '{tag}': native comm_core available "
+              f"here but not on every rank; all ranks using torch.distributed",
+              file=sys.stderr, flush=True)
+    if all_native:
+        backend = RcclBackend(torch.device("cuda",
+                                           torch.cuda.current_device()), tag)
+        PROVENANCE[tag] = "rccl-native"
+        return backend
     if torch.cuda.is_available() and dist.get_backend() != "gloo" and \
-            os.environ.get("DEAR_FORCE_TORCH_COMM", "0") != "1":
-        try:
-            return RcclBackend(torch.device("cuda",
-                                            torch.cuda.current_device()), tag)
-        except Exception as e:  # noqa: BLE001
-            # availability valve: keep the job alive on its torch.distributed
-            # nccl(=RCCL) channels, but NEVER silently — tests/smoke exercise
-            # the native path directly and fail loudly there.
-            import sys
-            print(f"[dear] WARNING: native comm_core channel '{tag}' failed "
-                  f"({type(e).__name__}: {e}); falling back to "
-                  f"torch.distributed process-group channels",
-                  file=sys.stderr, flush=True)
-            if os.environ.get("DEAR_STRICT_NATIVE_COMM", "0") == "1":
-                raise
+            os.environ.get("DEAR_STRICT_NATIVE_COMM", "0") == "1":
+        raise RuntimeError(
+            f"DEAR_STRICT_NATIVE_COMM=1 but native comm_core unavailable for "
+            f"channel '{tag}' on at least one rank")
     # independent PG per channel so RS/AG are separate traffic streams
     group = dist.new_group(backend=dist.get_backend())
+    PROVENANCE[tag] = "torch-dist"
     return TorchDistBackend(group)

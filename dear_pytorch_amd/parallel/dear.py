@@ -110,6 +110,7 @@ class DearOptimizer(torch.optim.Optimizer):
         self._init_sched_state()
         if self.rank == 0 and os.environ.get("DEAR_QUIET", "0") != "1":
             import sys
+            n = len(self.groups)
             mb = [g.nbytes / 1e6 for g in self.groups]
             print(f"[dear] {n} fusion groups, sizes MB: "
                   f"{', '.join(f'{m:.1f}' for m in mb)}", file=sys.stderr,
