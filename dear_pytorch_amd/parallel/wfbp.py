@@ -113,9 +113,14 @@ class WfbpOptimizer(torch.optim.Optimizer):
         k = vals.numel()
         av = torch.empty(k * self.size, device=vals.device, dtype=vals.dtype)
         ai = torch.empty(k * self.size, device=idx.device, dtype=idx.dtype)
-        h1 = self.backend.all_gather(vals.contiguous(), av)
-        h2 = self.backend.all_gather(idx.contiguous(), ai)
-        self._sparse_ctx[group.index] = (av, ai)
+        vals_c, idx_c = vals.contiguous(), idx.contiguous()
+        h1 = self.backend.all_gather(vals_c, av)
+        h2 = self.backend.all_gather(idx_c, ai)
+        # keep the SEND tensors alive until step() has waited on the handles:
+        # the side-stream RCCL read races the caching allocator otherwise
+        # (ADVICE r1 — freeing vals/idx here lets their storage be reused
+        # while the comm stream is still reading).
+        self._sparse_ctx[group.index] = (av, ai, vals_c, idx_c)
         self._handles[group.index] = (h1, h2)
 
     def zero_grad(self, set_to_none: bool = False):
@@ -132,7 +137,7 @@ class WfbpOptimizer(torch.optim.Optimizer):
                 h.wait_compute()
             self._handles[g.index] = NULL_HANDLE
             if g.index in self._sparse_ctx:
-                av, ai = self._sparse_ctx.pop(g.index)
+                av, ai, _vals_c, _idx_c = self._sparse_ctx.pop(g.index)
                 g.bucket.zero_()
                 g.bucket.scatter_add_(0, ai, av)
         if self.size > 1:

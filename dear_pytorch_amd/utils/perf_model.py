@@ -20,7 +20,8 @@ DEFAULT_ALPHA = 15e-6
 DEFAULT_BETA = 1.0 / (100e9)
 
 __all__ = ["AlphaBeta", "fit_alpha_beta", "predict_allreduce_time",
-           "plan_mgwfbp_flags", "topk_perf_model", "allgather_perf_model",
+           "plan_mgwfbp_flags", "mgwfbp_merge_plan", "mgwfbp_schedule_time",
+           "topk_perf_model", "allgather_perf_model",
            "predict_density_with_size_and_computation",
            "gen_threshold_from_normal_distribution"]
 
@@ -53,55 +54,128 @@ def predict_allreduce_time(nbytes: int, world: int,
     return (ab or AlphaBeta()).allreduce_time(nbytes, world)
 
 
+# Tensors below this many bytes always merge forward — a collective this
+# small is pure startup cost (reference force-merges <8192 elements).
+_TINY_BYTES = 8192 * 4
+
+
+def mgwfbp_merge_plan(sizes_bytes: Sequence[int], tb: Sequence[float],
+                      world: int, ab: Optional[AlphaBeta] = None) -> List[int]:
+    """Core MG-WFBP decision (pure function, backward execution order).
+
+    Inputs are per-layer payload bytes and backward-compute times, index 0 =
+    the FIRST layer to finish backward.  Returns a group id per layer (group
+    ids increase along backward order; layers in one group communicate as one
+    merged collective when the group's last layer is ready).
+
+    Model (reference __calculate_comm_start recurrence, wfbp/dopt.py:409-470):
+    all collectives share one comm channel and issue in backward order, so
+    comm of layer i starts at  start[i] = max(start[i-1] + t_c[i-1], ready[i]).
+    Merging layer i into i+1 removes one startup ``alpha`` but delays i's
+    payload until ready[i+1]; merge when the channel would otherwise sit idle
+    (start[i] >= ready[i+1]) or the idle gap it closes is < alpha.
+    """
+    ab = ab or AlphaBeta()
+    L = len(sizes_bytes)
+    if L == 0:
+        return []
+    p = [float(s) for s in sizes_bytes]
+    tc = [ab.allreduce_time(s, world) for s in p]
+    ready = [0.0] * L
+    acc = 0.0
+    for i in range(L):
+        acc += tb[i]
+        ready[i] = acc
+
+    def comm_starts():
+        start = [0.0] * L
+        start[0] = ready[0]
+        for i in range(1, L):
+            start[i] = max(start[i - 1] + tc[i - 1], ready[i])
+        return start
+
+    merged_next = [False] * L  # layer i joins layer i+1's group
+    for i in range(L - 1):
+        start = comm_starts()
+        nxt_ready = ready[i + 1]
+        do_merge = False
+        if nxt_ready < start[i] + tc[i]:
+            # i's comm would still be in flight when i+1 becomes ready
+            if start[i] >= nxt_ready:
+                do_merge = True       # comm hasn't even started: free merge
+            elif (nxt_ready - start[i]) < ab.alpha:
+                do_merge = True       # idle gap closed is cheaper than alpha
+        if not do_merge and p[i] < _TINY_BYTES:
+            do_merge = True           # startup-dominated tiny payload
+        if do_merge:
+            p[i + 1] += p[i]
+            p[i] = 0.0
+            tc[i] = 0.0
+            tc[i + 1] = ab.allreduce_time(p[i + 1], world)
+            merged_next[i] = True
+    gids = [0] * L
+    for i in range(1, L):
+        gids[i] = gids[i - 1] + (0 if merged_next[i - 1] else 1)
+    return gids
+
+
+def mgwfbp_schedule_time(sizes_bytes: Sequence[int], tb: Sequence[float],
+                         gids: Sequence[int], world: int,
+                         ab: Optional[AlphaBeta] = None) -> float:
+    """Predicted iteration span (backward start → last collective done) for a
+    given grouping, same serialized-channel model as mgwfbp_merge_plan.  Used
+    by the planner tests to check plans against exhaustive small-case optima.
+    """
+    ab = ab or AlphaBeta()
+    L = len(sizes_bytes)
+    if L == 0:
+        return 0.0
+    ready_layer = []
+    acc = 0.0
+    for i in range(L):
+        acc += tb[i]
+        ready_layer.append(acc)
+    # group payloads and ready times (a group is ready at its LAST layer)
+    n_groups = max(gids) + 1
+    g_bytes = [0.0] * n_groups
+    g_ready = [0.0] * n_groups
+    for i in range(L):
+        g_bytes[gids[i]] += sizes_bytes[i]
+        g_ready[gids[i]] = max(g_ready[gids[i]], ready_layer[i])
+    t = 0.0
+    for g in range(n_groups):
+        t = max(t, g_ready[g]) + ab.allreduce_time(g_bytes[g], world)
+    return t
+
+
 def plan_mgwfbp_flags(model: torch.nn.Module, backend=None,
                       layerwise_times: Optional[Dict[str, float]] = None,
                       ab: Optional[AlphaBeta] = None) -> List[bool]:
     """MG-WFBP merge plan → per-module 'start new group' flags (forward order).
 
-    Algorithm (capability of reference _generate_groups_mgwfbp): walk modules
-    in BACKWARD order with their measured backward-compute times tau_b; a
-    layer's collective can start when its gradients are ready; merge layer l
-    into the following group when the extra wait it causes is smaller than the
-    saved startup alpha.  Flags are broadcast from rank 0 so every rank builds
-    identical buckets.
+    Wraps mgwfbp_merge_plan (the reference's start-time recurrence,
+    wfbp/dopt.py:409-470) over the model's modules; flags are broadcast from
+    rank 0 so every rank builds identical buckets.
     """
     from ..parallel.fusion import _module_param_order
     mods = _module_param_order(model)
     n = len(mods)
     if n == 0:
         return []
-    ab = ab or AlphaBeta()
     world = backend.size if backend is not None else 1
     if layerwise_times is None:
         from ..profiling import Profiling
         layerwise_times = Profiling.estimate_backward_times(model)
     # backward order = reverse forward order
-    sizes = [sum(p.numel() for _, p in ps) * 4 for _, ps in mods]
-    taus = [layerwise_times.get(id(m), 1e-4) if isinstance(layerwise_times, dict)
-            else 1e-4 for m, _ in mods]
-    flags = [False] * n  # True = start new group at module i (forward order)
-    flags[0] = True
-    # Greedy backward-order merge: accumulate bytes; a new group starts (in
-    # backward order) when the accumulated communication would overlap worse
-    # than paying a fresh startup — i.e. when comm time for the merged group
-    # exceeds the backward compute time remaining to hide it by more than
-    # alpha.
-    acc_bytes = 0
-    remaining_tau = 0.0
-    for i in range(n - 1, 0, -1):  # backward order, boundary decided at i
-        acc_bytes += sizes[i]
-        remaining_tau += taus[i]
-        t_merged = ab.allreduce_time(acc_bytes + sizes[i - 1], world)
-        t_split = ab.allreduce_time(acc_bytes, world) + \
-            ab.allreduce_time(sizes[i - 1], world)
-        # force-merge tiny tensors (reference merges <8192 elems, :467)
-        if sizes[i - 1] < 8192 * 4:
-            continue
-        if t_merged > t_split - ab.alpha + remaining_tau:
-            flags[i] = True
-            acc_bytes = 0
-            remaining_tau = 0.0
-    if backend is not None and world > 1:
+    sizes_bw = [sum(p.numel() for _, p in ps) * 4 for _, ps in reversed(mods)]
+    taus_bw = [layerwise_times.get(id(m), 1e-4)
+               if isinstance(layerwise_times, dict) else 1e-4
+               for m, _ in reversed(mods)]
+    gids_bw = mgwfbp_merge_plan(sizes_bw, taus_bw, world, ab)
+    gid_fw = list(reversed(gids_bw))
+    flags = [True] + [gid_fw[i] != gid_fw[i - 1] for i in range(1, n)]
+    import torch.distributed as dist
+    if backend is not None and world > 1 and dist.is_initialized():
         from .dist_helpers import bcast_floats
         flags = [bool(v) for v in bcast_floats([1.0 if f else 0.0
                                                 for f in flags])]
