@@ -40,9 +40,15 @@ def parse_args():
     p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--method", default="dear",
                    choices=["dear", "dear-bo", "ddp", "wfbp", "mgwfbp",
-                            "naive", "rb", "bytescheduler"],
+                            "asc", "mgs", "naive", "rb", "bytescheduler"],
                    help="gradient-sync method (dear is the product)")
     p.add_argument("--threshold-mb", type=float, default=25.0)
+    p.add_argument("--compressor", default="none",
+                   choices=["none", "topk", "eftopk", "gaussian", "sign",
+                            "efsign"],
+                   help="gradient compression codec (wfbp-family methods)")
+    p.add_argument("--density", type=float, default=1.0,
+                   help="kept gradient fraction for sparse codecs")
     p.add_argument("--exclude-parts", default="")
     p.add_argument("--no-fusion", action="store_true")
     p.add_argument("--comm-dtype", default="fp32",
@@ -160,8 +166,12 @@ def wrap_method(args, model, opt_fn):
             return model, _TunedStep()
         return model, opt
     from dear_pytorch_amd.parallel import baselines
+    kw = {}
+    if args.method in ("wfbp", "mgwfbp", "asc", "mgs") and \
+            args.compressor != "none" and args.density < 1.0:
+        kw = dict(compressor=args.compressor, density=args.density)
     opt = baselines.make(args.method, opt_fn(model.parameters()), model,
-                         threshold_bytes=threshold)
+                         threshold_bytes=threshold, **kw)
     return model, opt
 
 
@@ -274,6 +284,8 @@ def main():
                 "parallelism": f"dp{world}",
                 "method": args.method,
                 "threshold_mb": None if args.no_fusion else args.threshold_mb,
+                "compressor": None if args.compressor == "none"
+                else f"{args.compressor}@{args.density}",
                 "comm_backend": _provenance_str(args),
             },
         }

@@ -30,7 +30,11 @@ def main():
     p.add_argument("--num-iters", type=int, default=5)
     p.add_argument("--method", default="dear",
                    choices=["dear", "dear-bo", "dear-wt", "ddp", "wfbp", "mgwfbp",
-                            "naive", "rb", "bytescheduler"])
+                            "asc", "mgs", "naive", "rb", "bytescheduler"])
+    p.add_argument("--asc", action="store_true",
+                   help="alias for --method asc (reference --asc flag)")
+    p.add_argument("--mgs", action="store_true",
+                   help="alias for --method mgs (reference MGS-SGD planning)")
     p.add_argument("--threshold", type=int, default=25 * 1024 * 1024,
                    help="fusion threshold bytes")
     p.add_argument("--no-fusion", action="store_true")
@@ -42,6 +46,10 @@ def main():
                    help="bf16 autocast compute (reference --fp16 axis; "
                         "gradients/optimizer stay fp32)")
     args = p.parse_args()
+    if args.asc:
+        args.method = "asc"
+    if args.mgs:
+        args.method = "mgs"
     if args.timeline:
         os.environ["DEAR_TIMELINE"] = args.timeline  # chrome-trace of the
         # DeAR queue events (reference WFSGD_TIMELINE via horovod_mpi_cj.sh)
@@ -103,10 +111,10 @@ def main():
     else:
         from dear_pytorch_amd.parallel import baselines
         kw = {}
-        if args.method in ("wfbp", "mgwfbp") and args.compressor != "none" \
-                and args.density < 1.0:
+        if args.method in ("wfbp", "mgwfbp", "asc", "mgs") and \
+                args.compressor != "none" and args.density < 1.0:
             kw = dict(compressor=args.compressor, density=args.density)
-        if args.method == "mgwfbp":
+        if args.method in ("mgwfbp", "asc", "mgs"):
             # reference protocol: measure per-layer backward times first and
             # hand them to the planner (mgwfbp/imagenet_benchmark.py:98-100)
             from dear_pytorch_amd.profiling import Profiling

@@ -25,7 +25,19 @@ def make(method: str, optimizer, model, threshold_bytes=None, **kw):
                              **kw)
     if method == "mgwfbp":
         return WfbpOptimizer(optimizer, model, threshold_bytes=threshold_bytes,
-                             mgwfbp=True, **kw)
+                             mode="mgwfbp", **kw)
+    if method == "asc":
+        # merge only when the comm channel would idle (reference
+        # _generate_groups_asc, hv_distributed_optimizer.py:353-428)
+        return WfbpOptimizer(optimizer, model, threshold_bytes=threshold_bytes,
+                             mode="asc", **kw)
+    if method == "mgs":
+        # merged gradient sparsification: top-k + sparse allgather planning
+        # (reference _generate_groups_mgs, hv_distributed_optimizer.py:430-508)
+        kw.setdefault("compressor", "topk")
+        kw.setdefault("density", 0.01)
+        return WfbpOptimizer(optimizer, model, threshold_bytes=threshold_bytes,
+                             mode="mgs", **kw)
     if method == "naive":
         return NaiveDearOptimizer(optimizer, model, **kw)
     if method == "bytescheduler":

@@ -26,6 +26,7 @@ __all__ = ["WfbpOptimizer"]
 class WfbpOptimizer(torch.optim.Optimizer):
     def __init__(self, optimizer: torch.optim.Optimizer, model: torch.nn.Module,
                  threshold_bytes: Optional[int] = None, mgwfbp: bool = False,
+                 mode: Optional[str] = None,
                  fusion_flags: Optional[list] = None,
                  backend: Optional[CommBackend] = None,
                  compressor: Optional[str] = None, density: float = 1.0,
@@ -35,7 +36,15 @@ class WfbpOptimizer(torch.optim.Optimizer):
         self.backend = backend or create_backend("wfbp")
         self.rank, self.size = self.backend.rank, self.backend.size
         self._device = next(model.parameters()).device
-        self._mgwfbp = mgwfbp
+        # merge-planner mode: None (threshold/per-layer), 'mgwfbp' (start-time
+        # recurrence, merge when wait < alpha), 'asc' (merge only when the
+        # channel would idle anyway), 'mgs' (sparse-allgather-aware planning,
+        # reference hv_distributed_optimizer.py:353-508)
+        if mode is None and mgwfbp:
+            mode = "mgwfbp"
+        assert mode in (None, "mgwfbp", "asc", "mgs"), mode
+        self.mode = mode
+        self._mgwfbp = mode == "mgwfbp"
         self._layerwise_times = layerwise_times
         # sparsified sync (reference wfbp/dopt.py compression machinery):
         # top-k style codecs communicate (values, indices) via all-gather
@@ -45,8 +54,8 @@ class WfbpOptimizer(torch.optim.Optimizer):
             from ..compression import compressors
             self.compressor = compressors[compressor]()
         self._sparse_ctx = {}
-        if mgwfbp and fusion_flags is None:
-            fusion_flags = self._plan_mgwfbp()
+        if mode is not None and fusion_flags is None:
+            fusion_flags = self._plan_flags(mode)
         self.groups: List[BucketGroup] = build_groups(
             model, threshold_bytes, fusion_flags=fusion_flags)
         for g in self.groups:
@@ -65,12 +74,12 @@ class WfbpOptimizer(torch.optim.Optimizer):
                 acc.register_hook(self._make_hook(p))
                 self._grad_accs.append(acc)
 
-    def _plan_mgwfbp(self):
-        """Measure per-layer backward times + fit xGMI alpha-beta, then merge
-        layers where waiting is cheaper than the saved startup alpha
-        (reference _generate_groups_mgwfbp, wfbp/dopt.py:380-486, with
+    def _plan_flags(self, mode):
+        """Measure per-layer backward times + fit xGMI alpha-beta, then run the
+        selected merge planner (reference _generate_groups_{mgwfbp,asc,mgs},
+        wfbp/dopt.py:380-486 / hv_distributed_optimizer.py:353-508, with
         MEASURED constants instead of the Ethernet tables)."""
-        from ..utils.perf_model import plan_mgwfbp_flags
+        from ..utils import perf_model as pm
         ab = None
         if self.size > 1 and self._device.type == "cuda":
             # fit alpha-beta on the live xGMI fabric (reference fits with
@@ -79,11 +88,21 @@ class WfbpOptimizer(torch.optim.Optimizer):
             ab = CommunicationProfiler(self.backend, iters=5).fit()
             if self.rank == 0:
                 import sys
-                print(f"[mgwfbp] measured alpha={ab.alpha * 1e6:.1f}us "
+                print(f"[{mode}] measured alpha={ab.alpha * 1e6:.1f}us "
                       f"beta={1.0 / ab.beta / 1e9:.1f}GB/s", file=sys.stderr,
                       flush=True)
-        return plan_mgwfbp_flags(self.model, self.backend,
-                                 layerwise_times=self._layerwise_times, ab=ab)
+        if mode == "asc":
+            return pm.plan_asc_flags(self.model, self.backend,
+                                     layerwise_times=self._layerwise_times,
+                                     ab=ab)
+        if mode == "mgs":
+            return pm.plan_mgs_flags(self.model, self.backend,
+                                     layerwise_times=self._layerwise_times,
+                                     density=self.density if self.density < 1.0
+                                     else 0.01, ab=ab)
+        return pm.plan_mgwfbp_flags(self.model, self.backend,
+                                    layerwise_times=self._layerwise_times,
+                                    ab=ab)
 
     def _make_hook(self, p):
         def hook(*_):

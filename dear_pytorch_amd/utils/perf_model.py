@@ -21,6 +21,8 @@ DEFAULT_BETA = 1.0 / (100e9)
 
 __all__ = ["AlphaBeta", "fit_alpha_beta", "predict_allreduce_time",
            "plan_mgwfbp_flags", "mgwfbp_merge_plan", "mgwfbp_schedule_time",
+           "plan_asc_flags", "asc_merge_plan",
+           "plan_mgs_flags", "mgs_merge_plan", "topk_compute_time",
            "topk_perf_model", "allgather_perf_model",
            "predict_density_with_size_and_computation",
            "gen_threshold_from_normal_distribution"]
@@ -148,15 +150,116 @@ def mgwfbp_schedule_time(sizes_bytes: Sequence[int], tb: Sequence[float],
     return t
 
 
-def plan_mgwfbp_flags(model: torch.nn.Module, backend=None,
-                      layerwise_times: Optional[Dict[str, float]] = None,
-                      ab: Optional[AlphaBeta] = None) -> List[bool]:
-    """MG-WFBP merge plan → per-module 'start new group' flags (forward order).
-
-    Wraps mgwfbp_merge_plan (the reference's start-time recurrence,
-    wfbp/dopt.py:409-470) over the model's modules; flags are broadcast from
-    rank 0 so every rank builds identical buckets.
+def asc_merge_plan(sizes_bytes: Sequence[int], tb: Sequence[float],
+                   world: int, ab: Optional[AlphaBeta] = None) -> List[int]:
+    """ASC plan (reference _generate_groups_asc, hv_distributed_optimizer.py
+    in mgwfbp/): same serialized-channel model as MG-WFBP but merge layer i
+    into i+1 ONLY when i's collective could not have started before i+1 is
+    ready anyway (a strictly free merge) — no alpha-saving merges and no
+    tiny-tensor force merge.  Backward execution order, index 0 first.
     """
+    ab = ab or AlphaBeta()
+    L = len(sizes_bytes)
+    if L == 0:
+        return []
+    p = [float(s) for s in sizes_bytes]
+    tc = [ab.allreduce_time(s, world) for s in p]
+    ready = [0.0] * L
+    acc = 0.0
+    for i in range(L):
+        acc += tb[i]
+        ready[i] = acc
+    merged_next = [False] * L
+    for i in range(L - 1):
+        start = [0.0] * L
+        start[0] = ready[0]
+        for j in range(1, L):
+            start[j] = max(start[j - 1] + tc[j - 1], ready[j])
+        if start[i] > ready[i + 1]:   # channel idle past i+1's readiness
+            p[i + 1] += p[i]
+            p[i] = 0.0
+            tc[i] = 0.0
+            tc[i + 1] = ab.allreduce_time(p[i + 1], world)
+            merged_next[i] = True
+    gids = [0] * L
+    for i in range(1, L):
+        gids[i] = gids[i - 1] + (0 if merged_next[i - 1] else 1)
+    return gids
+
+
+# fitted top-k kernel constant: t = TOPK_S * n * log2(n) (reference utils.py
+# topk_perf_model; refit on MI355X via CommunicationProfiler if needed)
+TOPK_S = 2.5e-10
+
+
+def topk_compute_time(n_elems: float, s: float = TOPK_S) -> float:
+    import math
+    if n_elems <= 1:
+        return 0.0
+    return s * n_elems * math.log2(n_elems)
+
+
+def mgs_merge_plan(sizes_elems: Sequence[int], tb: Sequence[float],
+                   world: int, density: float,
+                   ab: Optional[AlphaBeta] = None,
+                   topk_s: float = TOPK_S) -> List[int]:
+    """MGS-SGD plan (reference _generate_groups_mgs): sparsified sync where
+    each group is top-k compressed on the compute stream right after its
+    backward, then its (values, indices) are all-gathered on the comm channel.
+    Merge layer i into i+1 when the extra wait tw (deferred backward + the
+    merged top-k's superlinear cost − the channel idle gap that exists anyway)
+    is smaller than the allgather startup saved tsave.  Backward order,
+    index 0 first; 12 bytes/selected element (fp32 value + int64 index).
+    """
+    ab = ab or AlphaBeta()
+    L = len(sizes_elems)
+    if L == 0:
+        return []
+    p = [float(s) for s in sizes_elems]
+    tbl = [float(t) for t in tb]
+
+    def tk(n):
+        return topk_compute_time(n, topk_s)
+
+    def ag(n):
+        if n <= 0:
+            return 0.0
+        return allgather_perf_model(int(max(n * density, 1)) * 12, world, ab)
+
+    merged_next = [False] * L
+    for i in range(L - 1):
+        # compute-stream schedule: backward(i) then topk(i), serialized
+        sparse_done = [0.0] * L
+        t = 0.0
+        for j in range(L):
+            t += tbl[j] + tk(p[j])
+            sparse_done[j] = t
+        # comm channel start recurrence over allgathers
+        start = [0.0] * L
+        tc = [ag(p[j]) for j in range(L)]
+        start[0] = sparse_done[0]
+        for j in range(1, L):
+            start[j] = max(start[j - 1] + tc[j - 1], sparse_done[j])
+        idle_gap = start[i] - sparse_done[i]
+        tw = tbl[i + 1] + tk(p[i] + p[i + 1]) - tk(p[i]) - tk(p[i + 1]) \
+            - idle_gap
+        tsave = ag(p[i]) + ag(p[i + 1]) - ag(p[i] + p[i + 1])
+        if tw < tsave:
+            p[i + 1] += p[i]
+            p[i] = 0.0
+            tbl[i + 1] += tbl[i]
+            tbl[i] = 0.0
+            merged_next[i] = True
+    gids = [0] * L
+    for i in range(1, L):
+        gids[i] = gids[i - 1] + (0 if merged_next[i - 1] else 1)
+    return gids
+
+
+def _module_plan(model, backend, layerwise_times, planner):
+    """Shared flag plumbing: run `planner(sizes_bw, taus_bw, world)` over the
+    model's modules in backward order, map the group ids back to forward-order
+    'start new group' flags, broadcast from rank 0."""
     from ..parallel.fusion import _module_param_order
     mods = _module_param_order(model)
     n = len(mods)
@@ -166,12 +269,11 @@ def plan_mgwfbp_flags(model: torch.nn.Module, backend=None,
     if layerwise_times is None:
         from ..profiling import Profiling
         layerwise_times = Profiling.estimate_backward_times(model)
-    # backward order = reverse forward order
-    sizes_bw = [sum(p.numel() for _, p in ps) * 4 for _, ps in reversed(mods)]
+    sizes_bw = [sum(p.numel() for _, p in ps) for _, ps in reversed(mods)]
     taus_bw = [layerwise_times.get(id(m), 1e-4)
                if isinstance(layerwise_times, dict) else 1e-4
                for m, _ in reversed(mods)]
-    gids_bw = mgwfbp_merge_plan(sizes_bw, taus_bw, world, ab)
+    gids_bw = planner(sizes_bw, taus_bw, world)
     gid_fw = list(reversed(gids_bw))
     flags = [True] + [gid_fw[i] != gid_fw[i - 1] for i in range(1, n)]
     import torch.distributed as dist
@@ -180,6 +282,38 @@ def plan_mgwfbp_flags(model: torch.nn.Module, backend=None,
         flags = [bool(v) for v in bcast_floats([1.0 if f else 0.0
                                                 for f in flags])]
     return flags
+
+
+def plan_asc_flags(model: torch.nn.Module, backend=None,
+                   layerwise_times: Optional[Dict[str, float]] = None,
+                   ab: Optional[AlphaBeta] = None) -> List[bool]:
+    """ASC merge plan → per-module 'start new group' flags (forward order)."""
+    return _module_plan(model, backend, layerwise_times,
+                        lambda s, t, w: asc_merge_plan([x * 4 for x in s],
+                                                       t, w, ab))
+
+
+def plan_mgs_flags(model: torch.nn.Module, backend=None,
+                   layerwise_times: Optional[Dict[str, float]] = None,
+                   density: float = 0.01,
+                   ab: Optional[AlphaBeta] = None) -> List[bool]:
+    """MGS merge plan (sparse allgather) → forward-order flags."""
+    return _module_plan(model, backend, layerwise_times,
+                        lambda s, t, w: mgs_merge_plan(s, t, w, density, ab))
+
+
+def plan_mgwfbp_flags(model: torch.nn.Module, backend=None,
+                      layerwise_times: Optional[Dict[str, float]] = None,
+                      ab: Optional[AlphaBeta] = None) -> List[bool]:
+    """MG-WFBP merge plan → per-module 'start new group' flags (forward order).
+
+    Wraps mgwfbp_merge_plan (the reference's start-time recurrence,
+    wfbp/dopt.py:409-470) over the model's modules; flags are broadcast from
+    rank 0 so every rank builds identical buckets.
+    """
+    return _module_plan(model, backend, layerwise_times,
+                        lambda s, t, w: mgwfbp_merge_plan([x * 4 for x in s],
+                                                          t, w, ab))
 
 
 # ---- closed-form models carried over for the sparsification planner --------

@@ -86,6 +86,77 @@ def test_plan_beats_trivial_baselines_and_near_optimal():
     assert worst_ratio < 1.05, worst_ratio
 
 
+def test_asc_merges_only_free_merges():
+    """ASC merges layer i into i+1 only when i's collective could not start
+    before i+1 is ready — with an always-busy channel it merges; with long
+    compute gaps it never merges (unlike MG-WFBP, no alpha-saving merges)."""
+    from dear_pytorch_amd.utils.perf_model import asc_merge_plan
+    # channel saturated: t_c >> tb => later layers pile up => merges happen
+    ab = AlphaBeta(alpha=1e-5, beta=1e-8)
+    sizes = [10 << 20] * 6
+    tb = [1e-4] * 6
+    gids = asc_merge_plan(sizes, tb, world=8, ab=ab)
+    assert max(gids) < 5, gids
+    # compute-bound: channel always drains before next layer ready => no merge
+    gids2 = asc_merge_plan(sizes, [10.0] * 6, world=8, ab=ab)
+    assert max(gids2) == 5, gids2
+    # distinguishing pair: idle gap 0.6s is < alpha (1s) => MG-WFBP merges to
+    # save the startup, ASC refuses (comm of layer 0 starts before layer 1 is
+    # ready, so the merge is not free)
+    ab3 = AlphaBeta(alpha=1.0, beta=1e-12)
+    sizes3 = [1 << 24] * 2
+    tb3 = [0.6, 0.6]
+    asc3 = asc_merge_plan(sizes3, tb3, world=4, ab=ab3)
+    mg3 = mgwfbp_merge_plan(sizes3, tb3, world=4, ab=ab3)
+    assert max(asc3) == 1, asc3                # ASC: split
+    assert max(mg3) == 0, mg3                  # MG-WFBP: merged
+
+
+def test_asc_plan_never_worse_than_split():
+    from dear_pytorch_amd.utils.perf_model import asc_merge_plan
+    rng = random.Random(3)
+    for _ in range(20):
+        L = rng.randint(4, 8)
+        world = rng.choice([2, 4, 8])
+        ab = AlphaBeta(alpha=10 ** rng.uniform(-6, -3),
+                       beta=10 ** rng.uniform(-12, -9))
+        sizes = [rng.choice([1 << 14, 1 << 18, 1 << 22]) for _ in range(L)]
+        tb = [10 ** rng.uniform(-5, -2) for _ in range(L)]
+        gids = asc_merge_plan(sizes, tb, world, ab)
+        t_plan = mgwfbp_schedule_time(sizes, tb, gids, world, ab)
+        t_split = mgwfbp_schedule_time(sizes, tb, list(range(L)), world, ab)
+        # free merges by construction cannot delay any group's start
+        assert t_plan <= t_split * (1 + 1e-9)
+
+
+def test_mgs_plan_merges_when_allgather_startup_dominates():
+    from dear_pytorch_amd.utils.perf_model import mgs_merge_plan
+    # huge alpha per allgather, cheap topk: merge everything
+    ab = AlphaBeta(alpha=5e-3, beta=1e-11)
+    sizes = [1 << 20] * 5
+    tb = [1e-5] * 5
+    gids = mgs_merge_plan(sizes, tb, world=8, density=0.01, ab=ab,
+                          topk_s=1e-12)
+    assert max(gids) == 0, gids
+    # expensive superlinear topk + negligible startup: keep split
+    ab2 = AlphaBeta(alpha=1e-9, beta=1e-12)
+    gids2 = mgs_merge_plan(sizes, [1e-3] * 5, world=8, density=0.01, ab=ab2,
+                           topk_s=1e-6)
+    assert max(gids2) == 4, gids2
+
+
+def test_plan_flag_wrappers_shapes():
+    import torch.nn as nn
+    from dear_pytorch_amd.utils.perf_model import plan_asc_flags, plan_mgs_flags
+    m = nn.Sequential(nn.Linear(64, 64), nn.ReLU(), nn.Linear(64, 64),
+                      nn.Linear(64, 8))
+    from dear_pytorch_amd.parallel.fusion import _module_param_order
+    n = len(_module_param_order(m))
+    for flags in (plan_asc_flags(m), plan_mgs_flags(m, density=0.05)):
+        assert len(flags) == n
+        assert flags[0]
+
+
 def test_flags_reflect_regime(cpu_model_factory=None):
     """End-to-end through plan_mgwfbp_flags: comm-bound alpha-beta yields one
     group; compute-bound with huge per-layer gaps yields several."""
