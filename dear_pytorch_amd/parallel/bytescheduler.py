@@ -4,22 +4,30 @@ Reference capability: the bytescheduler/ benchmark path (C23 in SURVEY.md —
 `bsc.ScheduledOptimizer(model, optimizer, max_steps)` over Horovod).  The
 external bytescheduler library negotiates op order through Horovod's
 coordinator; RCCL instead requires every rank to issue collectives in the
-same order, so this implementation uses a DETERMINISTIC priority drain:
+same order, so this implementation uses a DETERMINISTIC credit scheduler:
 
 * each parameter tensor is partitioned into <= partition_bytes chunks
   (ByteScheduler's tensor partitioning — small high-priority pieces are not
   stuck behind a large low-priority transfer),
 * ready chunks enter a priority queue keyed by FORWARD order (layers needed
   earliest next iteration sync first — ByteScheduler's priority rule),
-* the queue is drained in priority order at every hook/step boundary; since
-  the backward-ready sequence is identical on every rank, the issue order is
-  identical too (RCCL-safe by construction).
+* at every drain opportunity (each backward hook and step()) the queue is
+  drained in priority order, but only while the in-flight byte budget
+  (``credit_bytes``) lasts; chunks that do not fit stay QUEUED across hooks,
+  so a high-priority chunk arriving later (backward visits last layers
+  first) preempts queued low-priority chunks at the next opportunity,
+* credit is replenished by retiring the oldest in-flight transfer; since
+  retirement order, the backward-ready sequence and every scheduling
+  decision are functions of rank-identical state, all ranks issue the same
+  collective order (RCCL-safe by construction, no coordinator needed).
 
-step() waits for all chunks, averages, and applies the wrapped optimizer.
+step() drains with blocking credit waits, waits for all chunks, averages,
+and applies the wrapped optimizer.
 """
 from __future__ import annotations
 
 import heapq
+from collections import deque
 from typing import List, Optional
 
 import torch
@@ -32,6 +40,7 @@ __all__ = ["ByteSchedulerOptimizer"]
 
 class ByteSchedulerOptimizer(torch.optim.Optimizer):
     def __init__(self, optimizer, model, partition_bytes: int = 4 * 1024 * 1024,
+                 credit_bytes: Optional[int] = None,
                  backend: Optional[CommBackend] = None, **kw):
         self.optim = optimizer
         self.model = model
@@ -39,6 +48,10 @@ class ByteSchedulerOptimizer(torch.optim.Optimizer):
         self.rank, self.size = self.backend.rank, self.backend.size
         self._device = next(model.parameters()).device
         self.partition = max(partition_bytes // 4, 1024)  # elements
+        # in-flight budget: ByteScheduler's credit. Default 4 partitions —
+        # enough to keep the channel busy, small enough that priority matters.
+        self.credit_bytes = credit_bytes if credit_bytes is not None \
+            else 4 * self.partition * 4
         # one group per module; priority = forward position
         self.groups: List[BucketGroup] = build_groups(model, None)
         for g in self.groups:
@@ -48,9 +61,10 @@ class ByteSchedulerOptimizer(torch.optim.Optimizer):
             for s in g.slots:
                 self._slot_of[s.param] = (g, s)
         self._ready = [0] * len(self.groups)
-        self._heap = []  # (priority, seq, chunk_tensor)
+        self._heap = []  # (priority=fwd index, seq, chunk_tensor)
         self._seq = 0
-        self._handles = []
+        self._inflight = deque()  # (handle, nbytes), oldest first
+        self._outstanding = 0     # bytes issued, not yet retired
         self._grad_accs = []
         for g in self.groups:
             for s in g.slots:
@@ -71,7 +85,7 @@ class ByteSchedulerOptimizer(torch.optim.Optimizer):
             self._ready[group.index] += 1
             if self._ready[group.index] == len(group.slots):
                 self._enqueue(group)
-                self._drain()
+                self._drain(block=False)
         return hook
 
     def _enqueue(self, group: BucketGroup):
@@ -84,21 +98,44 @@ class ByteSchedulerOptimizer(torch.optim.Optimizer):
             self._seq += 1
             off = end
 
-    def _drain(self):
+    def _issue(self, chunk, nbytes):
+        if self.size > 1:
+            h = self.backend.all_reduce(chunk)
+            self._inflight.append((h, nbytes))
+            self._outstanding += nbytes
+
+    def _retire_oldest(self):
+        h, nb = self._inflight.popleft()
+        h.host_wait()
+        self._outstanding -= nb
+
+    def _drain(self, block: bool):
+        """Issue queued chunks in priority order while credit lasts.
+
+        block=False (backward hooks): stop when the budget is exhausted —
+        the remaining chunks stay queued so later, higher-priority arrivals
+        can overtake them.  block=True (step()): recover credit by retiring
+        the oldest in-flight transfer and keep going.
+        """
         while self._heap:
+            nbytes = self._heap[0][2].numel() * self._heap[0][2].element_size()
+            while self._outstanding + nbytes > self.credit_bytes \
+                    and self._inflight:
+                if not block:
+                    return
+                self._retire_oldest()
+            # a chunk larger than the whole budget issues alone
             _, _, chunk = heapq.heappop(self._heap)
-            if self.size > 1:
-                self._handles.append(self.backend.all_reduce(chunk))
+            self._issue(chunk, nbytes)
 
     def zero_grad(self, set_to_none: bool = False):
         pass
 
     def step(self, closure=None):
         loss = closure() if closure is not None else None
-        self._drain()
-        for h in self._handles:
-            h.wait_compute()
-        self._handles.clear()
+        self._drain(block=True)
+        while self._inflight:
+            self._retire_oldest()
         if self.size > 1:
             for g in self.groups:
                 g.bucket.mul_(1.0 / self.size)
