@@ -107,6 +107,41 @@ def _ensure_state(optim, group, names):
             group.extra["adam_step"] = int(st0["step"])
 
 
+def readopt_group_state(optim, group):
+    """After ``optim.load_state_dict``: fold the freshly loaded per-param
+    state back into the contiguous slabs and reinstall slab views, so the
+    fused kernels honor a mid-training resume (loaded state was
+    silently ignored once slabs existed)."""
+    key_map = {"momentum": "momentum_buffer", "exp_avg": "exp_avg",
+               "exp_avg_sq": "exp_avg_sq"}
+    for name, key in key_map.items():
+        slab = group.extra.get(name)
+        if slab is None:
+            continue
+        found = False
+        for s in group.slots:
+            st = optim.state.get(s.param)
+            if st and key in st and torch.is_tensor(st[key]):
+                view = slab[s.offset: s.offset + s.numel]
+                if st[key].data_ptr() != view.data_ptr():
+                    view.copy_(st[key].reshape(-1).to(slab.dtype,
+                                                      copy=False))
+                found = True
+        if name == "momentum":
+            if found:
+                group.extra["mom_init"] = True
+            else:
+                slab.zero_()          # checkpoint predates the first step
+                group.extra["mom_init"] = False
+    st0 = optim.state.get(group.slots[0].param) or {}
+    if "step" in st0 and "adam_step" in group.extra:
+        step = st0["step"]
+        group.extra["adam_step"] = int(step.item() if torch.is_tensor(step)
+                                       else step)
+    group.extra["state_synced"] = False   # reinstall views on next step
+    group.extra.pop("shadow_opt", None)   # shadow re-adopts lazily
+
+
 def detach_group_state(optim, group):
     """Before freeing a group (regroup): turn the optimizer-state views into
     standalone tensors so state survives the slab teardown."""
@@ -247,7 +282,13 @@ def _adam_python(group, hyp, scale, decoupled_wd, step):
 
 
 def _shadow_step(optim, group, scale):
-    """Generic path: per-group shadow optimizer of the same class."""
+    """Generic path: per-group shadow optimizer of the same class.
+
+    The shadow's per-param state dicts are SHARED with the wrapped
+    optimizer's ``state`` (same dict objects), so ``optim.state_dict()``
+    sees real state and a ``load_state_dict`` between steps is adopted on
+    the next step (shadow state used to be checkpoint-invisible).
+    """
     opt = group.extra.get("shadow_opt")
     if opt is None:
         pg_of = {}
@@ -261,10 +302,18 @@ def _shadow_step(optim, group, scale):
             groups.append({"params": [s.param], **h})
         opt = type(optim)(groups)
         group.extra["shadow_opt"] = opt
-    sub = group.bucket[: group.numel] if group.numel else group.bucket
+    # resume path: a load_state_dict on the wrapped optimizer installs fresh
+    # state dicts there — let them win over the shadow's
+    for s in group.slots:
+        wrapped = optim.state.get(s.param)
+        if wrapped is not None and len(wrapped) and \
+                wrapped is not opt.state.get(s.param):
+            opt.state[s.param] = wrapped
     group.bucket.mul_(scale)          # averaged grads visible through views
     opt.step()
     group.bucket.zero_()
+    for s in group.slots:             # same dict object: stays in sync
+        optim.state[s.param] = opt.state[s.param]
 
 
 # --------------------------------------------------------------------- native

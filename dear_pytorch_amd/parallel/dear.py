@@ -300,6 +300,10 @@ class DearOptimizer(torch.optim.Optimizer):
 
     def load_state_dict(self, sd):
         self.optim.load_state_dict(sd)
+        # fold loaded per-param state back into the fused slabs so resumed
+        # training actually uses it
+        for g in self.groups:
+            fused_ops.readopt_group_state(self.optim, g)
 
     def summary(self) -> str:
         """Human-readable fusion-plan table (group sizes, module spans, wire

@@ -143,6 +143,68 @@ def test_state_dict_roundtrip():
     opt.load_state_dict(sd)
 
 
+@pytest.mark.parametrize("name", ["sgd_mom", "adam", "rmsprop"])
+def test_midtraining_checkpoint_resume_matches(name):
+    """Save after 3 steps, load into a FRESH wrapper, continue 3 more steps:
+    the resumed run must track the uninterrupted one exactly (the
+    shadow/generic path lost state across state_dict/load_state_dict, and
+    load after the first step was ignored by the fused slabs)."""
+    data = _data(T=6)
+    # uninterrupted run
+    m_ref = _model()
+    opt_ref = dear.DistributedOptimizer(OPTS[name](m_ref.parameters()),
+                                        model=m_ref)
+    for x, y in data:
+        nn.functional.mse_loss(m_ref(x), y).backward()
+        opt_ref.step()
+    opt_ref.synchronize()
+    # interrupted at step 3
+    m1 = _model()
+    opt1 = dear.DistributedOptimizer(OPTS[name](m1.parameters()), model=m1)
+    for x, y in data[:3]:
+        nn.functional.mse_loss(m1(x), y).backward()
+        opt1.step()
+    opt1.synchronize()
+    osd = copy.deepcopy(opt1.state_dict())
+    msd = copy.deepcopy(m1.state_dict())
+    assert osd["state"], f"{name}: state_dict must not be empty"
+    # resume in a brand-new process-like wrapper
+    m2 = _model()
+    m2.load_state_dict(msd)
+    opt2 = dear.DistributedOptimizer(OPTS[name](m2.parameters()), model=m2)
+    # take one throwaway-free path: load AFTER construction (slabs may exist)
+    nn.functional.mse_loss(m2(data[0][0]), data[0][1]).backward()
+    opt2.step()
+    opt2.synchronize()
+    m2.load_state_dict(msd)
+    opt2.load_state_dict(osd)
+    for x, y in data[3:]:
+        nn.functional.mse_loss(m2(x), y).backward()
+        opt2.step()
+    opt2.synchronize()
+    for (na, pa), (_, pb) in zip(m_ref.named_parameters(),
+                                 m2.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), \
+            f"{name} {na}: {(pa - pb).abs().max().item():.3e}"
+
+
+def test_generic_optimizer_state_dict_not_empty():
+    """RMSprop runs through the shadow path; its square_avg state must be
+    visible in state_dict() after steps."""
+    m = _model()
+    opt = dear.DistributedOptimizer(
+        torch.optim.RMSprop(m.parameters(), lr=1e-3), model=m)
+    data = _data(T=2)
+    for x, y in data:
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+    opt.synchronize()
+    sd = opt.state_dict()
+    assert sd["state"]
+    any_sq = any("square_avg" in st for st in sd["state"].values())
+    assert any_sq, "RMSprop square_avg missing from checkpoint"
+
+
 def test_gradient_accumulation_matches_serial_big_batch():
     """accum_steps=2 with half batches == one step on the full batch (loss
     scaled by 1/accum so gradients average)."""
