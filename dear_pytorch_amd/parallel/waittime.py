@@ -6,6 +6,15 @@ module's gradients sit in the bucket before the group completes during
 backward, then split groups so no gradient waits longer than a cycle budget
 (reference CYCLE_TIME = 5 ms, dopt_rsag_wt.py:40; EMA alpha = 0.9; flags
 broadcast from rank 0 at a fixed warmup step for cross-rank consistency).
+
+On GPU the waits are DEVICE times: each push and each group completion
+records a hipEvent on the compute stream and the gaps are read back with
+``elapsed_time`` after the iteration's sync (host
+``perf_counter`` in backward hooks measures kernel-launch spacing, not how
+long gradients actually sit in the bucket — hooks run far ahead of the
+device).  The reference paid a full ``torch.cuda.synchronize()`` per hook
+instead (profiling.py:47); events are free at hook time and only one sync
+per measured iteration is needed.
 """
 from __future__ import annotations
 
@@ -27,7 +36,11 @@ class WaitTimeAdaptiveFusion:
         self.verbose = verbose and opt.rank == 0
         self._step = 0
         self._push_t: Dict[int, float] = {}
-        self._wait: Dict[int, float] = {}      # module-id -> EMA wait seconds
+        self._push_ev: Dict[int, torch.cuda.Event] = {}
+        self._pending = []                     # (group, done_event) per iter
+        self._wait: Dict[int, float] = {}      # param-id -> EMA wait seconds
+        self._on_gpu = (torch.cuda.is_available()
+                        and opt._device.type == "cuda")
         self._orig_hook = opt._make_bw_hook
         self._done = False
         self._install()
@@ -42,22 +55,27 @@ class WaitTimeAdaptiveFusion:
             inner = outer._orig_hook(p)
 
             def hook(*a):
-                outer._push_t[id(p)] = time.perf_counter()
+                if outer._on_gpu:
+                    ev = torch.cuda.Event(enable_timing=True)
+                    ev.record()
+                    outer._push_ev[id(p)] = ev
+                else:
+                    outer._push_t[id(p)] = time.perf_counter()
                 g, s = opt._slot_of[p]
                 pre = opt._ready_count[g.index]
                 inner(*a)
                 # the inner hook resets ready_count to 0 on completion
                 if pre == len(g.slots) - 1:
-                    done = time.perf_counter()
-                    for slot in g.slots:
-                        t0 = outer._push_t.get(id(slot.param))
-                        if t0 is None:
-                            continue
-                        w = done - t0
-                        mid = id(slot.param)
-                        prev = outer._wait.get(mid, w)
-                        outer._wait[mid] = outer.ema * prev + \
-                            (1 - outer.ema) * w
+                    if outer._on_gpu:
+                        done_ev = torch.cuda.Event(enable_timing=True)
+                        done_ev.record()
+                        outer._pending.append((g, done_ev))
+                    else:
+                        done = time.perf_counter()
+                        for slot in g.slots:
+                            t0 = outer._push_t.get(id(slot.param))
+                            if t0 is not None:
+                                outer._ema_update(id(slot.param), done - t0)
             return hook
 
         opt._make_bw_hook = make_hook
@@ -66,10 +84,33 @@ class WaitTimeAdaptiveFusion:
         opt._remove_hooks()
         opt._register_hooks()
 
+    def _ema_update(self, pid: int, w: float):
+        prev = self._wait.get(pid, w)
+        self._wait[pid] = self.ema * prev + (1 - self.ema) * w
+
+    def _harvest_events(self):
+        """Resolve this iteration's device-time waits (one sync, GPU only)."""
+        if not self._pending:
+            return
+        torch.cuda.synchronize()
+        for g, done_ev in self._pending:
+            for slot in g.slots:
+                ev = self._push_ev.get(id(slot.param))
+                if ev is None:
+                    continue
+                self._ema_update(id(slot.param),
+                                 ev.elapsed_time(done_ev) * 1e-3)
+        self._pending.clear()
+        self._push_ev.clear()
+
     def step_end(self):
         """Call once per training iteration, after opt.step()."""
         self._step += 1
-        if self._done or self._step < self.regroup_at:
+        if self._done:
+            return
+        if self._on_gpu:
+            self._harvest_events()
+        if self._step < self.regroup_at:
             return
         flags = self._flags_from_waits()
         flags = self._sync_flags(flags)

@@ -167,3 +167,57 @@ def test_resnet50_training_step_gpu():
     opt.synchronize()
     assert all(map(lambda v: v == v, losses))  # finite
     assert losses[-1] < losses[0]  # learning on the fixed batch
+
+
+def test_waittime_device_measurement_correlates_with_injected_delay():
+    """Wait-time adaptive fusion must measure DEVICE wait (hipEvents), not
+    host hook spacing: inject a ~20 ms device sleep in
+    backward between two layers; the parameter that becomes ready BEFORE the
+    sleep must show a wait >= the sleep, the one after ~0."""
+    import dear_pytorch_amd as dear
+    from dear_pytorch_amd.parallel.waittime import WaitTimeAdaptiveFusion
+
+    # calibrate torch.cuda._sleep cycles for ~20 ms
+    torch.cuda.synchronize()
+    e0, e1 = (torch.cuda.Event(enable_timing=True) for _ in range(2))
+    e0.record(); torch.cuda._sleep(1 << 22); e1.record()
+    torch.cuda.synchronize()
+    per_cycle_ms = e0.elapsed_time(e1) / float(1 << 22)
+    cycles = max(int(20.0 / per_cycle_ms), 1)
+
+    class _BwdDelay(torch.autograd.Function):
+        @staticmethod
+        def forward(ctx, x):
+            return x
+
+        @staticmethod
+        def backward(ctx, g):
+            torch.cuda._sleep(cycles)
+            return g
+
+    class Delay(nn.Module):
+        def forward(self, x):
+            return _BwdDelay.apply(x)
+
+    torch.manual_seed(0)
+    m = nn.Sequential(nn.Linear(64, 64), Delay(), nn.Linear(64, 8)).to(_dev())
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.01), model=m,
+        threshold_bytes=1 << 30)  # one merged group
+    wt = WaitTimeAdaptiveFusion(opt, cycle_time_s=1.0, regroup_at_step=100,
+                                verbose=False)
+    x = torch.randn(16, 64, device=_dev())
+    y = torch.randn(16, 8, device=_dev())
+    for _ in range(3):
+        opt.zero_grad()
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+        wt.step_end()
+    opt.synchronize()
+    # backward order: layer2 (ready first, then the sleep), layer0 completes
+    last_lin = m[2]
+    first_lin = m[0]
+    w_early = max(wt._wait[id(p)] for p in last_lin.parameters())
+    w_late = max(wt._wait[id(p)] for p in first_lin.parameters())
+    assert w_early > 0.015, f"device wait not seen: {w_early * 1e3:.2f} ms"
+    assert w_late < w_early * 0.5, (w_late, w_early)
