@@ -45,6 +45,14 @@ def init(backend: str | None = None, timeout_s: int = 1800):
         return
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
+    # Single-node runs rendezvous on loopback (the driver/torchrun use
+    # 127.0.0.1).  RCCL's own TCP bootstrap (inside ncclCommInitRank) excludes
+    # loopback by default and can HANG on boxes whose other interfaces don't
+    # route between local processes — pin it to lo for loopback rendezvous.
+    # Data still moves over xGMI/P2P; the socket is bootstrap-only.
+    addr = os.environ.get("MASTER_ADDR", "127.0.0.1")
+    if addr in ("127.0.0.1", "localhost", "::1"):
+        os.environ.setdefault("NCCL_SOCKET_IFNAME", "lo")
     if torch.cuda.is_available():
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
     dist.init_process_group(backend=backend,

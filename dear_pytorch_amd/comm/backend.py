@@ -170,6 +170,16 @@ class TorchDistBackend(CommBackend):
         self.rank = dist.get_rank()
         self.size = dist.get_world_size()
         self._native_rs = dist.get_backend() == "nccl"
+        self._gloo = dist.get_backend() == "gloo"
+
+    def _stage(self, t):
+        """gloo + CUDA tensor: stage through the host (gloo's device support
+        is build-dependent; host staging makes multi-rank-on-one-GPU tests
+        deterministic). Returns (op_tensor, copy_back_or_None)."""
+        if self._gloo and t.is_cuda:
+            h = t.detach().cpu()
+            return h, (lambda: t.copy_(h))
+        return t, None
 
     def reduce_scatter(self, bucket, shard):
         if self._native_rs:
@@ -177,11 +187,14 @@ class TorchDistBackend(CommBackend):
                                               group=self.group, async_op=True)
             return _WorkHandle(work)
         n = shard.numel()
-        work = dist.all_reduce(bucket, op=dist.ReduceOp.SUM, group=self.group,
+        b, _ = self._stage(bucket)
+        work = dist.all_reduce(b, op=dist.ReduceOp.SUM, group=self.group,
                                async_op=True)
         r = self.rank
 
         def post():
+            if b is not bucket:
+                bucket.copy_(b)
             shard.copy_(bucket[r * n:(r + 1) * n])
 
         return _WorkHandle(work, post)
@@ -189,29 +202,43 @@ class TorchDistBackend(CommBackend):
     def all_gather(self, shard, bucket, after=None):
         if after is not None:
             after.wait_compute()
-        work = dist.all_gather_into_tensor(bucket, shard, group=self.group,
+        s, _ = self._stage(shard)
+        b = torch.empty_like(bucket, device=s.device) if s is not shard \
+            else bucket
+        work = dist.all_gather_into_tensor(b, s, group=self.group,
                                            async_op=True)
-        return _WorkHandle(work)
+        post = (lambda: bucket.copy_(b)) if b is not bucket else None
+        return _WorkHandle(work, post)
 
     def all_reduce(self, t):
-        return _WorkHandle(dist.all_reduce(t, group=self.group, async_op=True))
+        h, back = self._stage(t)
+        work = dist.all_reduce(h, group=self.group, async_op=True)
+        return _WorkHandle(work, back)
 
     def reduce(self, t, root):
-        return _WorkHandle(dist.reduce(t, root, group=self.group, async_op=True))
+        h, back = self._stage(t)
+        work = dist.reduce(h, root, group=self.group, async_op=True)
+        return _WorkHandle(work, back)
 
     def broadcast(self, t, root):
-        return _WorkHandle(dist.broadcast(t, root, group=self.group, async_op=True))
+        h, back = self._stage(t)
+        work = dist.broadcast(h, root, group=self.group, async_op=True)
+        return _WorkHandle(work, back)
 
     def send_recv(self, send, recv, peer):
+        s, _ = self._stage(send)
+        r, back = self._stage(recv)
         reqs = dist.batch_isend_irecv([
-            dist.P2POp(dist.isend, send, peer, group=self.group),
-            dist.P2POp(dist.irecv, recv, peer, group=self.group),
+            dist.P2POp(dist.isend, s, peer, group=self.group),
+            dist.P2POp(dist.irecv, r, peer, group=self.group),
         ])
 
         class _H(_Handle):
             def wait_compute(self):
-                for r in reqs:
-                    r.wait()
+                for req in reqs:
+                    req.wait()
+                if back is not None:
+                    back()
 
             host_wait = wait_compute
 
