@@ -41,7 +41,7 @@ def main():
     dear.init()
     rank, world = dear.rank(), dear.size()
     on_gpu = torch.cuda.is_available()
-    device = torch.device("cuda", dear.local_rank()) if on_gpu else "cpu"
+    device = dear.local_device()
     if on_gpu:
         torch.cuda.set_device(device)
 

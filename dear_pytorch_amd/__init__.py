@@ -54,7 +54,10 @@ def init(backend: str | None = None, timeout_s: int = 1800):
     if addr in ("127.0.0.1", "localhost", "::1"):
         os.environ.setdefault("NCCL_SOCKET_IFNAME", "lo")
     if torch.cuda.is_available():
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+        # modulo pinning (reference imagenet_benchmark.py:65 rank%4): lets
+        # world>device_count test configs share a GPU instead of crashing
+        lr = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(lr % torch.cuda.device_count())
     dist.init_process_group(backend=backend,
                             timeout=datetime.timedelta(seconds=timeout_s))
     _generic_backend = None  # created lazily after init
@@ -81,6 +84,13 @@ def size() -> int:
 
 def local_rank() -> int:
     return int(os.environ.get("LOCAL_RANK", "0"))
+
+
+def local_device() -> torch.device:
+    """This rank's pinned device (LOCAL_RANK modulo visible GPUs)."""
+    if torch.cuda.is_available():
+        return torch.device("cuda", local_rank() % torch.cuda.device_count())
+    return torch.device("cpu")
 
 
 def barrier():

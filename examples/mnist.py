@@ -53,7 +53,7 @@ def main():
     dear.init()
     rank, world = dear.rank(), dear.size()
     use_gpu = torch.cuda.is_available()
-    device = torch.device("cuda", dear.local_rank()) if use_gpu else "cpu"
+    device = dear.local_device()
     torch.manual_seed(42)
 
     train_ds = synthetic_mnist(args.train_size, seed=1)

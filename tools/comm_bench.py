@@ -58,8 +58,7 @@ def main():
 
     dear.init()
     rank, world = dear.rank(), dear.size()
-    device = torch.device("cuda", dear.local_rank()) \
-        if torch.cuda.is_available() else torch.device("cpu")
+    device = dear.local_device()
     if device.type == "cuda":
         torch.cuda.set_device(device)
     be = create_backend("commbench")
