@@ -223,3 +223,24 @@ def test_waittime_device_measurement_correlates_with_injected_delay():
     w_late = max(wt._wait[id(p)] for p in first_lin.parameters())
     assert w_early > 0.010, f"device wait not seen: {w_early * 1e3:.2f} ms"
     assert w_late < w_early * 0.5, (w_late, w_early)
+
+
+def test_topk_abs_native_bert_scale():
+    """Device top-k on BERT-Large bucket sizes: the native
+    count_ge/select_ge threshold-refine path must return exactly k abs-top
+    elements and agree with torch.topk on magnitude."""
+    from dear_pytorch_amd.ops.topk import topk_abs_native
+    dev = _dev()
+    torch.manual_seed(3)
+    for n, k in [(31_254_528, 312_545),   # BERT-Large-ish 125 MB bucket, 1%
+                 (6_553_600, 65_536),     # 25 MB bucket
+                 (1_048_576, 524_288)]:   # high density 50%
+        x = torch.randn(n, device=dev)
+        idx, val = topk_abs_native(x, k)
+        assert idx.numel() == k and val.numel() == k
+        assert torch.equal(x[idx], val)
+        ref_vals, _ = torch.topk(x.abs(), k)
+        # same magnitude multiset boundary: smallest selected |v| must be >=
+        # the k-th largest |x| minus ties tolerance
+        assert val.abs().min() >= ref_vals.min() - 1e-6
+        assert torch.allclose(val.abs().sum(), ref_vals.sum(), rtol=1e-4)
