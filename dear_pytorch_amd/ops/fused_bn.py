@@ -96,14 +96,28 @@ class FusedBNAct2d(nn.BatchNorm2d):
         act(bn(x) + residual)   with act = ReLU if relu else identity
     """
 
+    # (C, H*W) shapes where the MIOpen composition measured FASTER than the
+    # fused kernels in fwd+bwd (profiles/README.md microbench: C512 28^2 and
+    # C1024 14^2 lose ~0.04/0.02 ms per instance in backward) — dispatch
+    # empirically per shape.  Override with DEAR_BN_SKIP="C:HW,C:HW" or "none".
+    _skip_shapes = frozenset({(512, 28 * 28), (1024, 14 * 14)})
+
     def __init__(self, num_features, relu=False, **kw):
         super().__init__(num_features, **kw)
         self.relu = relu
         self._ws = {}  # rows -> _Workspace (variable batch sizes alternate)
+        import os
+        env = os.environ.get("DEAR_BN_SKIP")
+        if env is not None:
+            self._skip_shapes = frozenset(
+                () if env in ("", "none") else
+                (tuple(map(int, p.split(":"))) for p in env.split(",")))
 
     def _fast_ok(self, x, residual):
         return (x.is_cuda and x.dtype == torch.float32
                 and x.shape[1] % 4 == 0
+                and (x.shape[1], x.shape[2] * x.shape[3])
+                not in self._skip_shapes
                 and self.affine and self.track_running_stats
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and (residual is None or
