@@ -184,11 +184,16 @@ __global__ __launch_bounds__(kThreads) void bn_fwd_apply_kernel(
 }
 
 // ---- backward statistics: dy_eff, partial Σdy_eff and Σdy_eff·x̂ -----------
-template <bool kRelu, bool kStoreDyEff>
+// kRecomp (relu, no residual): the ReLU mask is recomputed from the
+// pre-activation fma(x, scale, shift) — bitwise-identical to the forward's —
+// instead of reading y, eliminating one full HBM pass over the tensor here
+// and another in bn_bwd_dx (7 -> 5 passes for the no-res ReLU BNs).
+template <bool kRelu, bool kStoreDyEff, bool kRecomp>
 __global__ __launch_bounds__(kThreads) void bn_bwd_stats_kernel(
     const float4* __restrict__ x, const float4* __restrict__ dy,
     const float4* __restrict__ y, float4* __restrict__ dy_eff,
     const float4* __restrict__ mean, const float4* __restrict__ invstd,
+    const float4* __restrict__ w, const float4* __restrict__ bp,
     long rows, int C4, Geom g, float4* __restrict__ pdb,
     float4* __restrict__ pdg) {
   __shared__ float4 ldb[kThreads], ldg[kThreads];
@@ -197,20 +202,35 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_stats_kernel(
   float4 sdb = make_float4(0, 0, 0, 0), sdg = make_float4(0, 0, 0, 0);
   if (act) {
     const float4 mc = mean[c4], ic = invstd[c4];
+    float4 scale, shift;
+    if (kRecomp) {
+      const float4 wc = w[c4], bc = bp[c4];
+      scale.x = ic.x * wc.x; shift.x = fmaf(-mc.x, scale.x, bc.x);
+      scale.y = ic.y * wc.y; shift.y = fmaf(-mc.y, scale.y, bc.y);
+      scale.z = ic.z * wc.z; shift.z = fmaf(-mc.z, scale.z, bc.z);
+      scale.w = ic.w * wc.w; shift.w = fmaf(-mc.w, scale.w, bc.w);
+    }
     long r0, r1;
     row_range(rows, g.rpb, r0, r1);
     for (long r = r0 + rsub; r < r1; r += g.rpb) {
       const long i = r * C4 + c4;
       float4 gg = dy[i];
+      const float4 xx = x[i];
       if (kRelu) {
-        const float4 yy = y[i];
-        gg.x = yy.x > 0.f ? gg.x : 0.f;
-        gg.y = yy.y > 0.f ? gg.y : 0.f;
-        gg.z = yy.z > 0.f ? gg.z : 0.f;
-        gg.w = yy.w > 0.f ? gg.w : 0.f;
+        if (kRecomp) {
+          gg.x = fmaf(xx.x, scale.x, shift.x) > 0.f ? gg.x : 0.f;
+          gg.y = fmaf(xx.y, scale.y, shift.y) > 0.f ? gg.y : 0.f;
+          gg.z = fmaf(xx.z, scale.z, shift.z) > 0.f ? gg.z : 0.f;
+          gg.w = fmaf(xx.w, scale.w, shift.w) > 0.f ? gg.w : 0.f;
+        } else {
+          const float4 yy = y[i];
+          gg.x = yy.x > 0.f ? gg.x : 0.f;
+          gg.y = yy.y > 0.f ? gg.y : 0.f;
+          gg.z = yy.z > 0.f ? gg.z : 0.f;
+          gg.w = yy.w > 0.f ? gg.w : 0.f;
+        }
       }
       if (kStoreDyEff) dy_eff[i] = gg;
-      const float4 xx = x[i];
       sdb = sdb + gg;
       sdg.x = fmaf(gg.x, (xx.x - mc.x) * ic.x, sdg.x);
       sdg.y = fmaf(gg.y, (xx.y - mc.y) * ic.y, sdg.y);
@@ -265,12 +285,13 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_reduce_kernel(
 }
 
 // ---- backward dx: dx = w*invstd*(dy_eff - Σdb/M - x̂·Σdg/M) ----------------
-template <bool kRelu, bool kHaveDyEff>
+template <bool kRelu, bool kHaveDyEff, bool kRecomp>
 __global__ __launch_bounds__(kThreads) void bn_bwd_dx_kernel(
     const float4* __restrict__ x, const float4* __restrict__ dy,
     const float4* __restrict__ y, const float4* __restrict__ dy_eff,
     const float4* __restrict__ mean, const float4* __restrict__ invstd,
-    const float4* __restrict__ w, const float4* __restrict__ dbeta,
+    const float4* __restrict__ w, const float4* __restrict__ bp,
+    const float4* __restrict__ dbeta,
     const float4* __restrict__ dgamma, long rows, int C4, Geom g,
     float4* __restrict__ dx) {
   int c4, rsub;
@@ -278,21 +299,35 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_dx_kernel(
   const float4 mc = mean[c4], ic = invstd[c4], wc = w[c4];
   const float inv_m = 1.f / (float)rows;
   const float4 db = dbeta[c4], dg = dgamma[c4];
-  float4 k, mdb, mdg;
+  float4 k, mdb, mdg, shift;
   k.x = ic.x * wc.x; mdb.x = db.x * inv_m; mdg.x = dg.x * inv_m;
   k.y = ic.y * wc.y; mdb.y = db.y * inv_m; mdg.y = dg.y * inv_m;
   k.z = ic.z * wc.z; mdb.z = db.z * inv_m; mdg.z = dg.z * inv_m;
   k.w = ic.w * wc.w; mdb.w = db.w * inv_m; mdg.w = dg.w * inv_m;
+  if (kRecomp) {
+    const float4 bc = bp[c4];
+    // k == the forward's scale; shift identical to the forward's
+    shift.x = fmaf(-mc.x, k.x, bc.x);
+    shift.y = fmaf(-mc.y, k.y, bc.y);
+    shift.z = fmaf(-mc.z, k.z, bc.z);
+    shift.w = fmaf(-mc.w, k.w, bc.w);
+  }
   long r0, r1;
   row_range(rows, g.rpb, r0, r1);
   for (long r = r0 + rsub; r < r1; r += g.rpb) {
     const long i = r * C4 + c4;
+    const float4 xx = x[i];
     float4 gg;
     if (kHaveDyEff) {
       gg = dy_eff[i];
     } else {
       gg = dy[i];
-      if (kRelu) {
+      if (kRelu && kRecomp) {
+        gg.x = fmaf(xx.x, k.x, shift.x) > 0.f ? gg.x : 0.f;
+        gg.y = fmaf(xx.y, k.y, shift.y) > 0.f ? gg.y : 0.f;
+        gg.z = fmaf(xx.z, k.z, shift.z) > 0.f ? gg.z : 0.f;
+        gg.w = fmaf(xx.w, k.w, shift.w) > 0.f ? gg.w : 0.f;
+      } else if (kRelu) {
         const float4 yy = y[i];
         gg.x = yy.x > 0.f ? gg.x : 0.f;
         gg.y = yy.y > 0.f ? gg.y : 0.f;
@@ -300,7 +335,6 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_dx_kernel(
         gg.w = yy.w > 0.f ? gg.w : 0.f;
       }
     }
-    const float4 xx = x[i];
     float4 o;
     o.x = k.x * (gg.x - mdb.x - (xx.x - mc.x) * ic.x * mdg.x);
     o.y = k.y * (gg.y - mdb.y - (xx.y - mc.y) * ic.y * mdg.y);
@@ -386,40 +420,42 @@ void dear_bn_fwd(hipStream_t stream, const float* x, const float* res,
 
 void dear_bn_bwd(hipStream_t stream, const float* x, const float* dy,
                  const float* y, float* dy_eff, const float* w,
-                 const float* mean, const float* invstd, float* pdb,
-                 float* pdg, int nparts, float* dbeta, float* dgamma,
-                 float* dx, long rows, int C, int relu, int want_dy_eff) {
+                 const float* b, const float* mean, const float* invstd,
+                 float* pdb, float* pdg, int nparts, float* dbeta,
+                 float* dgamma, float* dx, long rows, int C, int relu,
+                 int want_dy_eff) {
   const int C4 = C / 4;
   LaunchCfg cfg = make_cfg(rows, C4, nparts);
-#define STATS(R, S)                                                           \
-  hipLaunchKernelGGL((bn_bwd_stats_kernel<R, S>),                             \
+#define STATS(R, S, RC)                                                       \
+  hipLaunchKernelGGL((bn_bwd_stats_kernel<R, S, RC>),                         \
                      dim3(cfg.stats_rb, cfg.cblocks), dim3(kThreads), 0,      \
                      stream,                                                  \
                      (const float4*)x, (const float4*)dy, (const float4*)y,   \
                      (float4*)dy_eff, (const float4*)mean,                    \
-                     (const float4*)invstd, rows, C4, cfg.g, (float4*)pdb,    \
+                     (const float4*)invstd, (const float4*)w,                 \
+                     (const float4*)b, rows, C4, cfg.g, (float4*)pdb,         \
                      (float4*)pdg)
-  if (relu && want_dy_eff) STATS(true, true);
-  else if (relu) STATS(true, false);
-  else if (want_dy_eff) STATS(false, true);
-  else STATS(false, false);
+  if (relu && want_dy_eff) STATS(true, true, false);
+  else if (relu) STATS(true, false, true);   // no-res ReLU: y never read
+  else if (want_dy_eff) STATS(false, true, false);
+  else STATS(false, false, false);
 #undef STATS
   const int cpb2 = C4 < 64 ? C4 : 64;
   const int crb = (C4 + cpb2 - 1) / cpb2;
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(crb), dim3(kThreads), 0,
                      stream, (const float4*)pdb, (const float4*)pdg,
                      cfg.stats_rb, C4, dbeta, dgamma, cpb2);
-#define DX(R, H)                                                              \
-  hipLaunchKernelGGL((bn_bwd_dx_kernel<R, H>), dim3(cfg.rb, cfg.cblocks),     \
+#define DX(R, H, RC)                                                          \
+  hipLaunchKernelGGL((bn_bwd_dx_kernel<R, H, RC>), dim3(cfg.rb, cfg.cblocks), \
                      dim3(kThreads), 0, stream, (const float4*)x,             \
                      (const float4*)dy, (const float4*)y,                     \
                      (const float4*)dy_eff, (const float4*)mean,              \
                      (const float4*)invstd, (const float4*)w,                 \
-                     (const float4*)dbeta, (const float4*)dgamma, rows, C4,   \
-                     cfg.g, (float4*)dx)
-  if (want_dy_eff) DX(false, true);
-  else if (relu) DX(true, false);
-  else DX(false, false);
+                     (const float4*)b, (const float4*)dbeta,                  \
+                     (const float4*)dgamma, rows, C4, cfg.g, (float4*)dx)
+  if (want_dy_eff) DX(false, true, false);
+  else if (relu) DX(true, false, true);      // no-res ReLU: y never read
+  else DX(false, false, false);
 #undef DX
 }
 

@@ -359,8 +359,8 @@ extern "C" void dear_bn_fwd(hipStream_t, const float*, const float*, float*,
                             float, float, int, int);
 extern "C" void dear_bn_bwd(hipStream_t, const float*, const float*,
                             const float*, float*, const float*, const float*,
-                            const float*, float*, float*, int, float*, float*,
-                            float*, long, int, int, int);
+                            const float*, const float*, float*, float*, int,
+                            float*, float*, float*, long, int, int, int);
 
 static void bn_fwd(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor y,
                    at::Tensor w, at::Tensor b, at::Tensor mean,
@@ -381,15 +381,16 @@ static void bn_fwd(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor y,
 
 static void bn_bwd(at::Tensor x, at::Tensor dy, at::Tensor y,
                    c10::optional<at::Tensor> dy_eff, at::Tensor w,
-                   at::Tensor mean, at::Tensor invstd, at::Tensor pdb,
-                   at::Tensor pdg, at::Tensor dbeta, at::Tensor dgamma,
-                   at::Tensor dx, int64_t rows, int64_t C, bool relu,
-                   bool want_dy_eff) {
+                   at::Tensor b, at::Tensor mean, at::Tensor invstd,
+                   at::Tensor pdb, at::Tensor pdg, at::Tensor dbeta,
+                   at::Tensor dgamma, at::Tensor dx, int64_t rows, int64_t C,
+                   bool relu, bool want_dy_eff) {
   auto stream = c10::hip::getCurrentHIPStream();
   dear_bn_bwd(stream.stream(), x.data_ptr<float>(), dy.data_ptr<float>(),
               y.data_ptr<float>(),
               dy_eff ? dy_eff->data_ptr<float>() : nullptr,
-              w.data_ptr<float>(), mean.data_ptr<float>(),
+              w.data_ptr<float>(), b.data_ptr<float>(),
+              mean.data_ptr<float>(),
               invstd.data_ptr<float>(), pdb.data_ptr<float>(),
               pdg.data_ptr<float>(), (int)pdb.size(0),
               dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),

@@ -63,7 +63,7 @@ class _FusedBNFn(torch.autograd.Function):
         # workspace tensors are overwritten by the next forward, which would
         # silently corrupt a backward that runs after it (multi-forward
         # patterns like gradient accumulation)
-        ctx.save_for_backward(x, y, weight, mean.clone(), invstd.clone())
+        ctx.save_for_backward(x, y, weight, bias, mean.clone(), invstd.clone())
         ctx.relu = relu
         ctx.has_res = residual is not None
         ctx.dims = (rows, C)
@@ -73,7 +73,7 @@ class _FusedBNFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         import dear_pytorch_amd._kernels as K
-        x, y, weight, mean, invstd = ctx.saved_tensors
+        x, y, weight, bias, mean, invstd = ctx.saved_tensors
         rows, C = ctx.dims
         ws = ctx.ws
         dy = dy.contiguous(memory_format=torch.channels_last)
@@ -82,8 +82,8 @@ class _FusedBNFn(torch.autograd.Function):
         dbeta = torch.empty_like(dgamma)
         dy_eff = torch.empty_like(x) if ctx.has_res else None
         # partial buffers reused from forward (forward's reduce consumed them)
-        K.bn_bwd(x, dy, y, dy_eff, weight, mean, invstd, ws.psum, ws.psumsq,
-                 dbeta, dgamma, dx, rows, C, ctx.relu, ctx.has_res)
+        K.bn_bwd(x, dy, y, dy_eff, weight, bias, mean, invstd, ws.psum,
+                 ws.psumsq, dbeta, dgamma, dx, rows, C, ctx.relu, ctx.has_res)
         d_res = dy_eff if ctx.has_res else None
         return (dx, dgamma, dbeta, None, None, None, None, None, None, d_res,
                 None)
