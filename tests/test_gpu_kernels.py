@@ -236,7 +236,7 @@ def test_topk_abs_native_bert_scale():
                  (6_553_600, 65_536),     # 25 MB bucket
                  (1_048_576, 524_288)]:   # high density 50%
         x = torch.randn(n, device=dev)
-        idx, val = topk_abs_native(x, k)
+        val, idx = topk_abs_native(x, k)
         assert idx.numel() == k and val.numel() == k
         assert torch.equal(x[idx], val)
         ref_vals, _ = torch.topk(x.abs(), k)
