@@ -198,6 +198,32 @@ __global__ __launch_bounds__(kBlock) void pack_kernel(
   }
 }
 
+// += variant for packed-grad mode (dear.py): autograd ASSIGNS fresh grad
+// tensors (p.grad=None before backward — no per-param CUDAFunctor_add), and
+// one pack_add per bucket group folds them into the pre-zeroed bucket.
+// Accumulates so multi-micro-batch gradient accumulation still sums.
+__global__ __launch_bounds__(kBlock) void pack_add_kernel(
+    const int64_t* __restrict__ desc, int nchunks,
+    float* __restrict__ bucket) {
+  const int c = blockIdx.x;
+  if (c >= nchunks) return;
+  const float* __restrict__ src = reinterpret_cast<const float*>(desc[3 * c]);
+  float* __restrict__ dst = bucket + desc[3 * c + 1];
+  const int n = (int)desc[3 * c + 2];
+  if (aligned16(src) && aligned16(dst) && (n & 3) == 0) {
+    const float4* s4 = reinterpret_cast<const float4*>(src);
+    float4* d4 = reinterpret_cast<float4*>(dst);
+    for (int i = threadIdx.x; i < (n >> 2); i += kBlock) {
+      float4 a = d4[i];
+      const float4 b = s4[i];
+      a.x += b.x; a.y += b.y; a.z += b.z; a.w += b.w;
+      d4[i] = a;
+    }
+  } else {
+    for (int i = threadIdx.x; i < n; i += kBlock) dst[i] += src[i];
+  }
+}
+
 __global__ __launch_bounds__(kBlock) void unpack_scale_kernel(
     const int64_t* __restrict__ desc, int nchunks,
     const float* __restrict__ bucket, float scale) {
@@ -322,6 +348,15 @@ static void pack(at::Tensor desc, at::Tensor bucket) {
   HIP_CHECK(hipGetLastError());
 }
 
+static void pack_add(at::Tensor desc, at::Tensor bucket) {
+  const int nchunks = (int)desc.size(0);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(pack_add_kernel, dim3(nchunks), dim3(kBlock), 0,
+                     stream.stream(), desc.data_ptr<int64_t>(), nchunks,
+                     bucket.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+}
+
 static void unpack_scale(at::Tensor desc, at::Tensor bucket, double scale) {
   const int nchunks = (int)desc.size(0);
   auto stream = c10::hip::getCurrentHIPStream();
@@ -409,6 +444,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd);
   m.def("fused_adam", &fused_adam);
   m.def("pack", &pack);
+  m.def("pack_add", &pack_add);
   m.def("unpack_scale", &unpack_scale);
   m.def("count_ge", &count_ge);
   m.def("select_ge", &select_ge);

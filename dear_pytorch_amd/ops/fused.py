@@ -309,9 +309,20 @@ def _shadow_step(optim, group, scale):
         if wrapped is not None and len(wrapped) and \
                 wrapped is not opt.state.get(s.param):
             opt.state[s.param] = wrapped
+    # packed-grad mode keeps p.grad None; the shadow optimizer needs grads,
+    # so install bucket views for the step and restore None afterwards
+    from ..parallel.fusion import grad_view
+    restore_none = []
+    for s in group.slots:
+        if s.param.grad is None:
+            s.param.grad = grad_view(
+                group.bucket[s.offset: s.offset + s.numel], s.param)
+            restore_none.append(s.param)
     group.bucket.mul_(scale)          # averaged grads visible through views
     opt.step()
     group.bucket.zero_()
+    for p in restore_none:
+        p.grad = None
     for s in group.slots:             # same dict object: stays in sync
         optim.state[s.param] = opt.state[s.param]
 

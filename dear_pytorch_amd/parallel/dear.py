@@ -58,7 +58,8 @@ class DearOptimizer(torch.optim.Optimizer):
                  backend: Optional[CommBackend] = None,
                  comm_dtype: Optional[torch.dtype] = None,
                  accum_steps: int = 1,
-                 num_nearby_layers: Optional[int] = None):
+                 num_nearby_layers: Optional[int] = None,
+                 pack_grads: Optional[bool] = None):
         self.optim = optimizer
         self.model = model
         self.threshold_bytes = threshold_bytes
@@ -84,6 +85,14 @@ class DearOptimizer(torch.optim.Optimizer):
         self.rank, self.size = backend.rank, backend.size
 
         self._device = next(model.parameters()).device
+        # packed-grad mode (default on GPU): p.grad stays None so autograd
+        # ASSIGNS each gradient (no per-param accumulate-add kernel); one
+        # pack_add launch per bucket group folds them into the bucket.
+        # Measured on MI355X ResNet-50 bs64: ~161 CUDAFunctor_add kernels
+        # (~1.3 ms/step) collapse into 4 pack_add launches.
+        if pack_grads is None:
+            pack_grads = self._device.type == "cuda"
+        self.pack_grads = pack_grads
         self._num_steps = 0
         self._hook_handles = []
         self._grad_view_fixups = 0
@@ -101,7 +110,8 @@ class DearOptimizer(torch.optim.Optimizer):
             nearby_layers=self.num_nearby_layers)
         for g in self.groups:
             g.allocate(self.size, self._device,
-                       comm_dtype=self.comm_dtype if self.size > 1 else None)
+                       comm_dtype=self.comm_dtype if self.size > 1 else None,
+                       attach_grads=not self.pack_grads)
         self._param_group_of = {}
         self._slot_of = {}
         for g in self.groups:
@@ -160,18 +170,24 @@ class DearOptimizer(torch.optim.Optimizer):
     def _make_bw_hook(self, p):
         def hook(*_):
             group, slot = self._slot_of[p]
-            # grad-as-bucket-view safety: autograd may, in rare accumulation
-            # paths, replace .grad with a fresh tensor — detect and fold back.
-            bucket_slice = group.bucket[slot.offset: slot.offset + slot.numel]
-            if p.grad is not None and p.grad.data_ptr() != bucket_slice.data_ptr():
-                from .fusion import grad_view
-                gv = grad_view(bucket_slice, p)
-                gv.add_(p.grad)
-                p.grad = gv
-                self._grad_view_fixups += 1
+            if not self.pack_grads:
+                # grad-as-bucket-view safety: autograd may, in rare
+                # accumulation paths, replace .grad with a fresh tensor —
+                # detect and fold back.
+                bucket_slice = group.bucket[slot.offset:
+                                            slot.offset + slot.numel]
+                if p.grad is not None and \
+                        p.grad.data_ptr() != bucket_slice.data_ptr():
+                    from .fusion import grad_view
+                    gv = grad_view(bucket_slice, p)
+                    gv.add_(p.grad)
+                    p.grad = gv
+                    self._grad_view_fixups += 1
             self._ready_count[group.index] += 1
             if self._ready_count[group.index] == len(group.slots):
                 self._ready_count[group.index] = 0
+                if self.pack_grads:
+                    self._pack_group(group)
                 self._accum_count[group.index] += 1
                 if self._accum_count[group.index] >= self.accum_steps:
                     self._accum_count[group.index] = 0
@@ -180,6 +196,54 @@ class DearOptimizer(torch.optim.Optimizer):
                                              "comm")
                     self._launch_rs(group)
         return hook
+
+    def _pack_group(self, group: BucketGroup):
+        """Fold this backward's assigned grad tensors into the bucket with one
+        pack_add launch, then release them (bucket was zeroed by the fused
+        update, so += across micro-batches keeps accumulation semantics)."""
+        from .fusion import grad_view
+        if self._device.type != "cuda":
+            base = group.bucket.data_ptr()
+            for s in group.slots:
+                g = s.param.grad
+                if g is not None:
+                    if g.data_ptr() != base + 4 * s.offset:
+                        grad_view(group.bucket[s.offset: s.offset + s.numel],
+                                  s.param).add_(g)
+                    s.param.grad = None
+            return
+        rows = []
+        fallback = []
+        bdt = group.bucket.dtype
+        bucket_base = group.bucket.data_ptr()
+        for s in group.slots:
+            g = s.param.grad
+            if g is None:
+                continue
+            if g.data_ptr() == bucket_base + 4 * s.offset:
+                continue  # already IS the bucket slice (external view install)
+            # raw-pointer pack requires identical storage order; grads whose
+            # layout differs from the param's take the layout-aware view add
+            if g.dtype == bdt and (g.stride() == s.param.stride()
+                                   or (g.is_contiguous()
+                                       and s.param.is_contiguous())):
+                base = g.data_ptr()
+                off = 0
+                while off < s.numel:
+                    n = min(fused_ops.CHUNK, s.numel - off)
+                    rows.append((base + 4 * off, s.offset + off, n))
+                    off += n
+            else:
+                fallback.append(s)
+        if rows:
+            desc = torch.tensor(rows, dtype=torch.int64).to(
+                self._device, non_blocking=True)
+            fused_ops._native().pack_add(desc, group.bucket)
+        for s in fallback:
+            grad_view(group.bucket[s.offset: s.offset + s.numel],
+                      s.param).add_(s.param.grad)
+        for s in group.slots:
+            s.param.grad = None
 
     def _launch_rs(self, group: BucketGroup):
         if self.size > 1 and self._do_rs:

@@ -85,12 +85,16 @@ class BucketGroup:
         return self.padded * self.bucket.element_size() if self.bucket is not None else 0
 
     def allocate(self, world_size: int, device, dtype=torch.float32,
-                 comm_dtype=None):
+                 comm_dtype=None, attach_grads: bool = True):
         """Allocate the fused bucket + shard and point every param.grad at its
         slice.  With comm_dtype (bf16/fp16) a reduced-precision wire buffer is
         allocated alongside: collectives move half the xGMI bytes while
         accumulation stays fp32 (cast via one dtype-converting copy each way,
-        ~HBM-speed, negligible vs the comm saved)."""
+        ~HBM-speed, negligible vs the comm saved).
+
+        attach_grads=False (packed-grad mode, dear.py): grads stay None so
+        autograd ASSIGNS fresh tensors and one pack_add kernel per group folds
+        them into the bucket — instead of one add kernel per parameter."""
         self.padded = _align(max(self.numel, 1), ALIGN_ELEMS * world_size)
         self.bucket = torch.zeros(self.padded, device=device, dtype=dtype)
         shard_n = self.padded // world_size
@@ -102,7 +106,8 @@ class BucketGroup:
             self.shard = torch.empty(shard_n, device=device, dtype=dtype)
         for s in self.slots:
             s.param.grad = grad_view(
-                self.bucket[s.offset: s.offset + s.numel], s.param)
+                self.bucket[s.offset: s.offset + s.numel], s.param) \
+                if attach_grads else None
 
     def reattach_grads(self):
         """Re-point param.grad at bucket views (after anything detached them)."""

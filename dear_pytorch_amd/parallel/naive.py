@@ -29,6 +29,7 @@ class NaiveDearOptimizer(DearOptimizer):
         self.groups = groups
         for g in self.groups:
             g.allocate(self.size, self._device,
-                       comm_dtype=self.comm_dtype if self.size > 1 else None)
+                       comm_dtype=self.comm_dtype if self.size > 1 else None,
+                       attach_grads=not self.pack_grads)
         self._slot_of = {g.slots[0].param: (g, g.slots[0]) for g in self.groups}
         self._init_sched_state()
