@@ -92,6 +92,43 @@ __global__ __launch_bounds__(kThreads) void bn_fwd_stats_kernel(
   }
 }
 
+// ---- stage-A column reduce: [nparts][C4] partial pair -> [S][C4] ----------
+// The final reduce kernels run ONE block for small C (cblocks==1) and were
+// measured latency-bound at ~17 us; this deterministic pre-reduce splits the
+// partial dimension over blockIdx.y (chunk rows each) so the heavy part of
+// the reduction is parallel, leaving the single-block finalize <= 64 rows.
+constexpr int kRedChunk = 64;
+
+__global__ __launch_bounds__(kThreads) void col_reduce_pair_kernel(
+    const float4* __restrict__ a, const float4* __restrict__ b, int nparts,
+    int C4, int cpb2, float4* __restrict__ oa, float4* __restrict__ ob) {
+  __shared__ float4 l1[kThreads], l2[kThreads];
+  const int spb = kThreads / cpb2;
+  const int cl = threadIdx.x % cpb2;
+  const int sl = threadIdx.x / cpb2;
+  const int c4 = blockIdx.x * cpb2 + cl;
+  const int s0 = blockIdx.y * kRedChunk;
+  const int s1 = s0 + kRedChunk < nparts ? s0 + kRedChunk : nparts;
+  float4 s = make_float4(0, 0, 0, 0), ss = make_float4(0, 0, 0, 0);
+  if (c4 < C4 && sl < spb) {
+    for (int p = s0 + sl; p < s1; p += spb) {
+      s = s + a[(long)p * C4 + c4];
+      ss = ss + b[(long)p * C4 + c4];
+    }
+  }
+  l1[threadIdx.x] = s;
+  l2[threadIdx.x] = ss;
+  __syncthreads();
+  if (sl == 0 && c4 < C4) {
+    for (int k = 1; k < spb; ++k) {
+      s = s + l1[k * cpb2 + cl];
+      ss = ss + l2[k * cpb2 + cl];
+    }
+    oa[(long)blockIdx.y * C4 + c4] = s;
+    ob[(long)blockIdx.y * C4 + c4] = ss;
+  }
+}
+
 // ---- finalize mean / invstd + running stats -------------------------------
 // Cooperative column reduce of the [nparts][C4] float4 partial matrix:
 // a block covers cpb2 float4-columns x spb partial-slices (the v2 scalar
@@ -379,11 +416,31 @@ LaunchCfg make_cfg(long rows, int C4, int nparts_cap) {
 
 }  // namespace
 
+// two-stage reduce helper: pre-reduce [nparts][C4] pairs into the scratch
+// rows the python side allocates PAST the nparts partials (no extra buffers,
+// no binding change); returns the partial count the finalize kernel should
+// read and swaps the input pointers to the scratch when staged.
+static int pre_reduce(hipStream_t stream, int nparts, int C4,
+                      const float4*& a, const float4*& b, float4* sa,
+                      float4* sb) {
+  if (nparts <= kRedChunk) return nparts;
+  const int S = (nparts + kRedChunk - 1) / kRedChunk;
+  const int cpb2 = C4 < 64 ? C4 : 64;
+  const int crb = (C4 + cpb2 - 1) / cpb2;
+  hipLaunchKernelGGL(col_reduce_pair_kernel, dim3(crb, S), dim3(kThreads), 0,
+                     stream, a, b, nparts, C4, cpb2, sa, sb);
+  a = sa;
+  b = sb;
+  return S;
+}
+
 extern "C" {
 
 // host-visible helper so the python side can size partial buffers identically
+// (stats_rb rows + ceil(stats_rb/kRedChunk) stage-A scratch rows)
 int dear_bn_nparts(long rows, int C) {
-  return make_cfg(rows, C / 4, 0).stats_rb;
+  const int n = make_cfg(rows, C / 4, 0).stats_rb;
+  return n + (n + kRedChunk - 1) / kRedChunk;
 }
 
 void dear_bn_fwd(hipStream_t stream, const float* x, const float* res,
@@ -398,11 +455,16 @@ void dear_bn_fwd(hipStream_t stream, const float* x, const float* res,
                        dim3(cfg.stats_rb, cfg.cblocks), dim3(kThreads), 0,
                        stream, (const float4*)x, rows, C4, cfg.g,
                        (float4*)psum, (float4*)psumsq);
+    const float4* pa = (const float4*)psum;
+    const float4* pb = (const float4*)psumsq;
+    float4* sa = (float4*)psum + (size_t)cfg.stats_rb * C4;
+    float4* sb = (float4*)psumsq + (size_t)cfg.stats_rb * C4;
+    const int np = pre_reduce(stream, cfg.stats_rb, C4, pa, pb, sa, sb);
     const int cpb2 = C4 < 64 ? C4 : 64;
     const int crb = (C4 + cpb2 - 1) / cpb2;
     hipLaunchKernelGGL(bn_fwd_reduce_kernel, dim3(crb), dim3(kThreads), 0,
-                       stream, (const float4*)psum, (const float4*)psumsq,
-                       cfg.stats_rb, C4, rows, eps, momentum, mean, invstd,
+                       stream, pa, pb,
+                       np, C4, rows, eps, momentum, mean, invstd,
                        running_mean, running_var, 1, cpb2);
   }
 #define APPLY(R, S)                                                           \
@@ -440,11 +502,16 @@ void dear_bn_bwd(hipStream_t stream, const float* x, const float* dy,
   else if (want_dy_eff) STATS(false, true, false);
   else STATS(false, false, false);
 #undef STATS
+  const float4* pa = (const float4*)pdb;
+  const float4* pb = (const float4*)pdg;
+  float4* sa = (float4*)pdb + (size_t)cfg.stats_rb * C4;
+  float4* sb = (float4*)pdg + (size_t)cfg.stats_rb * C4;
+  const int np = pre_reduce(stream, cfg.stats_rb, C4, pa, pb, sa, sb);
   const int cpb2 = C4 < 64 ? C4 : 64;
   const int crb = (C4 + cpb2 - 1) / cpb2;
   hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(crb), dim3(kThreads), 0,
-                     stream, (const float4*)pdb, (const float4*)pdg,
-                     cfg.stats_rb, C4, dbeta, dgamma, cpb2);
+                     stream, pa, pb,
+                     np, C4, dbeta, dgamma, cpb2);
 #define DX(R, H, RC)                                                          \
   hipLaunchKernelGGL((bn_bwd_dx_kernel<R, H, RC>), dim3(cfg.rb, cfg.cblocks), \
                      dim3(kThreads), 0, stream, (const float4*)x,             \
