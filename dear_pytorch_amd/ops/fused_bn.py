@@ -96,11 +96,13 @@ class FusedBNAct2d(nn.BatchNorm2d):
         act(bn(x) + residual)   with act = ReLU if relu else identity
     """
 
-    # (C, H*W) shapes where the MIOpen composition measured FASTER than the
-    # fused kernels in fwd+bwd (profiles/README.md microbench: C512 28^2 and
-    # C1024 14^2 lose ~0.04/0.02 ms per instance in backward) — dispatch
-    # empirically per shape.  Override with DEAR_BN_SKIP="C:HW,C:HW" or "none".
-    _skip_shapes = frozenset({(512, 28 * 28), (1024, 14 * 14)})
+    # (C, H*W) shapes dispatched to the MIOpen composition instead of the
+    # fused kernels.  Empty since the two-stage partial reduce + recomputed
+    # ReLU mask made the fused path win on every ResNet-50 shape
+    # (profiles/README.md r2 history: with the r1 kernels C512@28^2 and
+    # C1024@14^2 lost in backward and were skipped here).  Override with
+    # DEAR_BN_SKIP="C:HW,C:HW" for empirical per-shape dispatch.
+    _skip_shapes = frozenset()
 
     def __init__(self, num_features, relu=False, **kw):
         super().__init__(num_features, **kw)
