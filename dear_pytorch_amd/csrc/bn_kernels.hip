@@ -249,8 +249,7 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_stats_kernel(
     }
     long r0, r1;
     row_range(rows, g.rpb, r0, r1);
-    for (long r = r0 + rsub; r < r1; r += g.rpb) {
-      const long i = r * C4 + c4;
+    auto body = [&](long i) {
       float4 gg = dy[i];
       const float4 xx = x[i];
       if (kRelu) {
@@ -273,7 +272,15 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_stats_kernel(
       sdg.y = fmaf(gg.y, (xx.y - mc.y) * ic.y, sdg.y);
       sdg.z = fmaf(gg.z, (xx.z - mc.z) * ic.z, sdg.z);
       sdg.w = fmaf(gg.w, (xx.w - mc.w) * ic.w, sdg.w);
+    };
+    // 2x row unroll (matches bn_fwd_stats): two float4 loads in flight per
+    // lane hides HBM latency at the low-occupancy small-spatial shapes
+    long r = r0 + rsub;
+    for (; r + g.rpb < r1; r += 2 * g.rpb) {
+      body(r * C4 + c4);
+      body((r + g.rpb) * C4 + c4);
     }
+    if (r < r1) body(r * C4 + c4);
   }
   ldb[threadIdx.x] = sdb;
   ldg[threadIdx.x] = sdg;
@@ -351,8 +358,7 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_dx_kernel(
   }
   long r0, r1;
   row_range(rows, g.rpb, r0, r1);
-  for (long r = r0 + rsub; r < r1; r += g.rpb) {
-    const long i = r * C4 + c4;
+  auto body = [&](long i) {
     const float4 xx = x[i];
     float4 gg;
     if (kHaveDyEff) {
@@ -378,7 +384,13 @@ __global__ __launch_bounds__(kThreads) void bn_bwd_dx_kernel(
     o.z = k.z * (gg.z - mdb.z - (xx.z - mc.z) * ic.z * mdg.z);
     o.w = k.w * (gg.w - mdb.w - (xx.w - mc.w) * ic.w * mdg.w);
     dx[i] = o;
+  };
+  long r = r0 + rsub;
+  for (; r + g.rpb < r1; r += 2 * g.rpb) {  // 2x unroll (see bwd_stats)
+    body(r * C4 + c4);
+    body((r + g.rpb) * C4 + c4);
   }
+  if (r < r1) body(r * C4 + c4);
 }
 
 struct LaunchCfg {
