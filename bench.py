@@ -102,7 +102,8 @@ def build_workload(args, device):
         bs = args.batch_size or 64
         fused_bn = args.fused_bn
         if fused_bn is None:
-            fused_bn = device.type == "cuda" and args.model.startswith("resnet")
+            fused_bn = device.type == "cuda" and args.model.startswith(
+                ("resnet", "densenet", "inception"))
         model = models.get_cnn(args.model, fused_bn=fused_bn).to(device)
         g = torch.Generator().manual_seed(1234)
         res = 299 if args.model == "inceptionv4" else 224

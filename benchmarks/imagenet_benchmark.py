@@ -69,7 +69,9 @@ def main():
         if rank == 0:
             print(msg, flush=True)
 
-    model = models.get_cnn(args.model).to(device)
+    fused_bn = on_gpu and args.model.startswith(("resnet", "densenet",
+                                                 "inception"))
+    model = models.get_cnn(args.model, fused_bn=fused_bn).to(device)
     res = 299 if args.model == "inceptionv4" else 224
     g = torch.Generator().manual_seed(55 + rank)
     data = torch.randn(args.batch_size, 3, res, res, generator=g).to(device)
