@@ -27,8 +27,10 @@ METHOD_SETS = {
     # with tensor fusion vs without (reference tf/notf sets)
     "tf": ["ddp", "mgwfbp", "dear"],
     "notf": ["wfbp", "naive", "dear-notf", "bytescheduler"],
-    "all": ["ddp", "wfbp", "mgwfbp", "naive", "rb", "bytescheduler", "dear",
-            "dear-bo"],
+    "all": ["ddp", "wfbp", "mgwfbp", "asc", "mgs", "naive", "rb",
+            "bytescheduler", "dear", "dear-bo", "dear-wt"],
+    # merge-planner comparison (mgwfbp vs asc vs mgs, reference hv_* modes)
+    "planners": ["mgwfbp", "asc", "mgs", "dear"],
     # time-breakdown ablations (reference dear/batch.sh exclude_parts sweeps)
     "breakdown": ["dear", "dear-nors", "dear-noag"],
 }
