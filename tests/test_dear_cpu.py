@@ -246,6 +246,48 @@ def test_generic_optimizer_state_dict_not_empty():
     assert any_sq, "RMSprop square_avg missing from checkpoint"
 
 
+def test_frozen_params_excluded_from_buckets():
+    """requires_grad=False params must not enter any bucket and training the
+    rest must match a serial run with the same freeze."""
+    def frozen_model():
+        m = _model()
+        for p in m[0].parameters():  # freeze first Linear
+            p.requires_grad_(False)
+        return m
+    data = _data(T=4)
+    a = frozen_model()
+    opt_a = torch.optim.SGD([p for p in a.parameters() if p.requires_grad],
+                            lr=0.05, momentum=0.9)
+    for x, y in data:
+        opt_a.zero_grad()
+        nn.functional.mse_loss(a(x), y).backward()
+        opt_a.step()
+    b = frozen_model()
+    opt_b = dear.DistributedOptimizer(
+        torch.optim.SGD([p for p in b.parameters() if p.requires_grad],
+                        lr=0.05, momentum=0.9), model=b)
+    frozen = {id(p) for p in b[0].parameters()}
+    for g in opt_b.groups:
+        for s in g.slots:
+            assert id(s.param) not in frozen
+    for x, y in data:
+        opt_b.zero_grad()
+        nn.functional.mse_loss(b(x), y).backward()
+        opt_b.step()
+    opt_b.synchronize()
+    for (na, pa), (_, pb) in zip(a.named_parameters(), b.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), na
+
+
+def test_pack_mode_with_comm_dtype_single_rank():
+    """pack mode + bf16 wire buffers coexist (wire cast only kicks in at
+    world>1; here just exercise the allocation path end-to-end)."""
+    data = _data(T=3)
+    m = _train_dear(_model(), OPTS["sgd_mom"], data, threshold_bytes=1 << 12,
+                    pack_grads=True, comm_dtype=torch.bfloat16)
+    assert all(torch.isfinite(p).all() for p in m.parameters())
+
+
 def test_gradient_accumulation_matches_serial_big_batch():
     """accum_steps=2 with half batches == one step on the full batch (loss
     scaled by 1/accum so gradients average)."""

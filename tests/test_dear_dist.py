@@ -64,6 +64,38 @@ def test_dear_ws2_matches_serial_full_batch():
         assert torch.equal(outs[0][k], outs[1][k])
 
 
+def _rank_train_pack(rank, world, T, bs):
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m,
+        threshold_bytes=1 << 12, pack_grads=True)
+    for x, y in _full_data(T, bs):
+        xs, ys = x[rank * bs:(rank + 1) * bs], y[rank * bs:(rank + 1) * bs]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    opt.synchronize()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_pack_mode_ws2_matches_serial():
+    """Packed-grad mode with real collectives (gloo ws2) == serial SGD."""
+    T, bs = 5, 8
+    ref = _serial_reference(T, bs)
+    outs = run_dist(_rank_train_pack, world_size=2, args=(T, bs))
+    for r, sd in enumerate(outs):
+        for k in ref:
+            assert torch.allclose(ref[k], sd[k], atol=1e-5), f"rank {r} {k}"
+    for k in ref:
+        assert torch.equal(outs[0][k], outs[1][k])
+
+
 def _rank_train_naive(rank, world, T, bs):
     import dear_pytorch_amd as dear
     from dear_pytorch_amd.parallel.naive import NaiveDearOptimizer
