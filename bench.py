@@ -181,7 +181,9 @@ def _provenance_str(args):
     contract: 'rccl-native', 'torch-dist' or 'local').
     DDP runs on torch's own ProcessGroupNCCL."""
     if args.method == "ddp":
-        return "ddp/torch-nccl"
+        import torch.distributed as dist
+        return "ddp/torch-" + (dist.get_backend() if dist.is_initialized()
+                               else "none")
     from dear_pytorch_amd.comm.backend import backend_provenance
     prov = backend_provenance()
     if not prov:
