@@ -39,8 +39,9 @@ def parse_args():
                    help="per-GPU batch (default: 64 CNN / 32 BERT)")
     p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--method", default="dear",
-                   choices=["dear", "dear-bo", "ddp", "wfbp", "mgwfbp",
-                            "asc", "mgs", "naive", "rb", "bytescheduler"],
+                   choices=["dear", "dear-bo", "dear-wt", "ddp", "wfbp",
+                            "mgwfbp", "asc", "mgs", "naive", "rb",
+                            "bytescheduler"],
                    help="gradient-sync method (dear is the product)")
     p.add_argument("--threshold-mb", type=float, default=25.0)
     p.add_argument("--compressor", default="none",
@@ -136,7 +137,7 @@ def wrap_method(args, model, opt_fn):
                 gradient_as_bucket_view=True)
         opt = opt_fn(model.parameters())
         return model, opt
-    if args.method in ("dear", "dear-bo"):
+    if args.method in ("dear", "dear-bo", "dear-wt"):
         cdt = {"fp32": None, "bf16": torch.bfloat16,
                "fp16": torch.float16}[args.comm_dtype]
         opt = dear.DistributedOptimizer(opt_fn(model.parameters()),
@@ -165,6 +166,21 @@ def wrap_method(args, model, opt_fn):
                     tuner.step_end()
                     return r
             return model, _TunedStep()
+        if args.method == "dear-wt":
+            from dear_pytorch_amd.parallel.waittime import \
+                WaitTimeAdaptiveFusion
+            wt = WaitTimeAdaptiveFusion(opt, verbose=False)
+            inner_wt = opt
+
+            class _WtStep:
+                def __getattr__(self, k):
+                    return getattr(inner_wt, k)
+
+                def step(self, *a, **k):
+                    r = inner_wt.step()
+                    wt.step_end()
+                    return r
+            return model, _WtStep()
         return model, opt
     from dear_pytorch_amd.parallel import baselines
     kw = {}
