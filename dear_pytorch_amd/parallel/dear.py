@@ -9,8 +9,10 @@ all-reduce is split into
     group's first forward-pre hook, pipelined with forward compute.
 
 MI355X-first differences from the reference design:
-  * gradients accumulate directly into the fused bucket (grad-as-bucket-view,
-    fusion.py) — no pack/unpack copies at all;
+  * packed-grad mode (GPU default): autograd ASSIGNS each gradient (p.grad
+    stays None, so no per-param accumulate-add kernels) and ONE pack_add
+    launch per bucket group folds them into the fused bucket; on CPU,
+    grad-as-bucket-view (fusion.py) accumulates in place instead;
   * RS and AG run on two dedicated RCCL communicators / HIP side streams with
     hipEvent dependency edges (compute→RS, RS→AG, AG→compute) instead of the
     reference's host-blocking ``synchronize()`` + placebo self-stream-wait
@@ -204,10 +206,11 @@ class DearOptimizer(torch.optim.Optimizer):
         from .fusion import grad_view
         if self._device.type != "cuda":
             base = group.bucket.data_ptr()
+            esz = group.bucket.element_size()
             for s in group.slots:
                 g = s.param.grad
                 if g is not None:
-                    if g.data_ptr() != base + 4 * s.offset:
+                    if g.data_ptr() != base + esz * s.offset:
                         grad_view(group.bucket[s.offset: s.offset + s.numel],
                                   s.param).add_(g)
                     s.param.grad = None
@@ -216,11 +219,12 @@ class DearOptimizer(torch.optim.Optimizer):
         fallback = []
         bdt = group.bucket.dtype
         bucket_base = group.bucket.data_ptr()
+        esz = group.bucket.element_size()
         for s in group.slots:
             g = s.param.grad
             if g is None:
                 continue
-            if g.data_ptr() == bucket_base + 4 * s.offset:
+            if g.data_ptr() == bucket_base + esz * s.offset:
                 continue  # already IS the bucket slice (external view install)
             # raw-pointer pack requires identical storage order; grads whose
             # layout differs from the param's take the layout-aware view add
@@ -231,7 +235,7 @@ class DearOptimizer(torch.optim.Optimizer):
                 off = 0
                 while off < s.numel:
                     n = min(fused_ops.CHUNK, s.numel - off)
-                    rows.append((base + 4 * off, s.offset + off, n))
+                    rows.append((base + esz * off, s.offset + off, n))
                     off += n
             else:
                 fallback.append(s)
