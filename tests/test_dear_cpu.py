@@ -246,6 +246,25 @@ def test_generic_optimizer_state_dict_not_empty():
     assert any_sq, "RMSprop square_avg missing from checkpoint"
 
 
+def test_pack_mode_regroup_between_iterations():
+    """BO-style mid-training regroup must work in packed-grad mode (buckets
+    freed and reallocated without grad views to restore)."""
+    data = _data(T=6)
+    a = _train_serial(_model(), OPTS["sgd_mom"], data)
+    m = _model()
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9,
+                        weight_decay=1e-4), model=m, pack_grads=True)
+    for i, (x, y) in enumerate(data):
+        nn.functional.mse_loss(m(x), y).backward()
+        opt.step()
+        if i == 2:
+            opt.regroup(1 << 12)
+    opt.synchronize()
+    for (na, pa), (_, pb) in zip(a.named_parameters(), m.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), na
+
+
 def test_frozen_params_excluded_from_buckets():
     """requires_grad=False params must not enter any bucket and training the
     rest must match a serial run with the same freeze."""

@@ -64,6 +64,43 @@ def test_dear_ws2_matches_serial_full_batch():
         assert torch.equal(outs[0][k], outs[1][k])
 
 
+def _rank_train_w(rank, world, T, bs):
+    # world-size-agnostic DeAR training on 1/world of the batch
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m,
+        threshold_bytes=1 << 12)
+    per = (2 * bs) // world
+    for x, y in _full_data(T, bs):
+        xs, ys = x[rank * per:(rank + 1) * per], y[rank * per:(rank + 1) * per]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    opt.synchronize()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_dear_ws4_matches_serial_full_batch():
+    """world_size=4 (padding/shard boundaries differ from ws2): still must
+    equal serial SGD on the combined batch — first line of defense for the
+    8-GPU day-one run."""
+    T, bs = 4, 8
+    ref = _serial_reference(T, bs)
+    outs = run_dist(_rank_train_w, world_size=4, args=(T, bs))
+    for r, sd in enumerate(outs):
+        for k in ref:
+            assert torch.allclose(ref[k], sd[k], atol=1e-5), f"rank {r} {k}"
+    for k in ref:
+        for r in range(1, 4):
+            assert torch.equal(outs[0][k], outs[r][k])
+
+
 def _rank_train_pack(rank, world, T, bs):
     import dear_pytorch_amd as dear
     dear.init(backend="gloo")
