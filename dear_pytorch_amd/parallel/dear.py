@@ -114,7 +114,6 @@ class DearOptimizer(torch.optim.Optimizer):
             g.allocate(self.size, self._device,
                        comm_dtype=self.comm_dtype if self.size > 1 else None,
                        attach_grads=not self.pack_grads)
-        self._param_group_of = {}
         self._slot_of = {}
         for g in self.groups:
             for s in g.slots:
