@@ -133,6 +133,29 @@ def test_pack_mode_ws2_matches_serial():
         assert torch.equal(outs[0][k], outs[1][k])
 
 
+def _rank_backend_agreement(rank, world):
+    """Rank 0 claims native capability, rank 1 does not: BOTH must agree on
+    torch-dist (a per-rank decision would leave rank 0 in ncclCommInitRank
+    while rank 1 sits in dist.new_group — ADVICE r1)."""
+    import dear_pytorch_amd as dear
+    from dear_pytorch_amd.comm import backend as B
+    dear.init(backend="gloo")
+    B._native_precheck = lambda: rank == 0  # simulate split capability
+    be = B.create_backend("agreement_test")
+    kind = type(be).__name__
+    prov = B.backend_provenance().get("agreement_test")
+    dear.shutdown()
+    return kind, prov
+
+
+@pytest.mark.timeout(300)
+def test_backend_choice_is_collective():
+    outs = run_dist(_rank_backend_agreement, world_size=2)
+    for kind, prov in outs:
+        assert kind == "TorchDistBackend", outs
+        assert prov == "torch-dist", outs
+
+
 def _rank_train_naive(rank, world, T, bs):
     import dear_pytorch_amd as dear
     from dear_pytorch_amd.parallel.naive import NaiveDearOptimizer
