@@ -85,19 +85,20 @@ def _rank_train_w(rank, world, T, bs):
     return out
 
 
-@pytest.mark.timeout(300)
-def test_dear_ws4_matches_serial_full_batch():
-    """world_size=4 (padding/shard boundaries differ from ws2): still must
-    equal serial SGD on the combined batch — first line of defense for the
-    8-GPU day-one run."""
+@pytest.mark.timeout(600)
+@pytest.mark.parametrize("world", [4, 8])
+def test_dear_wsN_matches_serial_full_batch(world):
+    """world_size 4 and 8 (padding/shard boundaries differ from ws2): still
+    must equal serial SGD on the combined batch — first line of defense for
+    the 8-GPU day-one run."""
     T, bs = 4, 8
     ref = _serial_reference(T, bs)
-    outs = run_dist(_rank_train_w, world_size=4, args=(T, bs))
+    outs = run_dist(_rank_train_w, world_size=world, args=(T, bs))
     for r, sd in enumerate(outs):
         for k in ref:
             assert torch.allclose(ref[k], sd[k], atol=1e-5), f"rank {r} {k}"
     for k in ref:
-        for r in range(1, 4):
+        for r in range(1, world):
             assert torch.equal(outs[0][k], outs[r][k])
 
 
