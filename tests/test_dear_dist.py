@@ -134,6 +134,38 @@ def test_pack_mode_ws2_matches_serial():
         assert torch.equal(outs[0][k], outs[1][k])
 
 
+def _rank_train_rb(rank, world, T, bs):
+    import dear_pytorch_amd as dear
+    from dear_pytorch_amd.parallel.rb import ReduceBcastOptimizer
+    dear.init(backend="gloo")
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = ReduceBcastOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), m,
+        threshold_bytes=1 << 12)
+    for x, y in _full_data(T, bs):
+        xs, ys = x[rank * bs:(rank + 1) * bs], y[rank * bs:(rank + 1) * bs]
+        opt.zero_grad()
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    opt.synchronize()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_rb_ws2_matches_serial():
+    """reduce+broadcast decomposition (reference dopt_rb.py) must equal
+    serial SGD on the combined batch."""
+    T, bs = 5, 8
+    ref = _serial_reference(T, bs)
+    outs = run_dist(_rank_train_rb, world_size=2, args=(T, bs))
+    for k in ref:
+        assert torch.allclose(ref[k], outs[0][k], atol=1e-5), k
+        assert torch.equal(outs[0][k], outs[1][k])
+
+
 def _rank_backend_agreement(rank, world):
     """Rank 0 claims native capability, rank 1 does not: BOTH must agree on
     torch-dist (a per-rank decision would leave rank 0 in ncclCommInitRank
