@@ -149,21 +149,23 @@ def broadcast_optimizer_state(optimizer, root_rank: int = 0):
     reference dopt_rsag.py:424-540."""
     if size() == 1:
         return
-    if isinstance(optimizer, DearOptimizer):
-        optimizer = optimizer.optim
+    # keep the wrapper so its load_state_dict can re-adopt the fused state
+    # slabs on receiving ranks (plain inner load would leave them stale)
+    wrapper = optimizer if isinstance(optimizer, DearOptimizer) else None
+    inner = optimizer.optim if wrapper is not None else optimizer
     # one-shot startup consistency: ship root's full state (scalars wrapped
     # with the tensors, mirroring the reference's tensor-wrap callbacks)
-    obj = [optimizer.state_dict() if rank() == root_rank else None]
+    obj = [inner.state_dict() if rank() == root_rank else None]
     dist.broadcast_object_list(obj, src=root_rank)
     if rank() != root_rank:
         cpu_state = obj[0]
         dev = next(iter(
-            p.device for g in optimizer.param_groups for p in g["params"]))
+            p.device for g in inner.param_groups for p in g["params"]))
         for st in cpu_state["state"].values():
             for k, v in st.items():
                 if torch.is_tensor(v) and v.dim() > 0:
                     st[k] = v.to(dev)
-        optimizer.load_state_dict(cpu_state)
+        (wrapper or inner).load_state_dict(cpu_state)
 
 
 # convenience namespace parity with the reference benchmark drivers

@@ -330,6 +330,45 @@ def test_broadcast_optimizer_state_ws2():
         assert abs(a[k] - b[k]) < 1e-6, k
 
 
+def _rank_bcast_wrapper_state(rank, world):
+    """broadcast_optimizer_state on the DearOptimizer WRAPPER must re-adopt
+    the fused slabs on receiving ranks (stale slabs would silently ignore
+    the broadcast and the ranks would diverge)."""
+    import dear_pytorch_amd as dear
+    dear.init(backend="gloo")
+    m = _model()
+    dear.broadcast_parameters(m.state_dict(), root_rank=0)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9), model=m,
+        threshold_bytes=1 << 12)
+    data = _full_data(6, 8)
+    for x, y in data[:2]:
+        xs, ys = x[rank * 8:(rank + 1) * 8], y[rank * 8:(rank + 1) * 8]
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    opt.synchronize()
+    if rank == 0:  # perturb root's momentum so the broadcast has to matter
+        for st in opt.optim.state.values():
+            if "momentum_buffer" in st:
+                st["momentum_buffer"].add_(0.5)
+    dear.broadcast_optimizer_state(opt, root_rank=0)
+    for x, y in data[2:]:
+        xs, ys = x[rank * 8:(rank + 1) * 8], y[rank * 8:(rank + 1) * 8]
+        nn.functional.mse_loss(m(xs), ys).backward()
+        opt.step()
+    opt.synchronize()
+    out = {k: v.clone() for k, v in m.state_dict().items()}
+    dear.shutdown()
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_broadcast_wrapper_state_readopts_slabs_ws2():
+    outs = run_dist(_rank_bcast_wrapper_state, world_size=2)
+    for k in outs[0]:
+        assert torch.equal(outs[0][k], outs[1][k]), k
+
+
 def _rank_train_bf16comm(rank, world, T, bs):
     import torch
     import dear_pytorch_amd as dear
