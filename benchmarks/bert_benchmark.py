@@ -25,7 +25,9 @@ def main():
     p.add_argument("--num-iters", type=int, default=5)
     p.add_argument("--method", default="dear",
                    choices=["dear", "dear-bo", "ddp", "wfbp", "mgwfbp",
-                            "naive", "rb", "bytescheduler"])
+                            "asc", "mgs", "naive", "rb", "bytescheduler"])
+    p.add_argument("--compressor", default="none")
+    p.add_argument("--density", type=float, default=1.0)
     p.add_argument("--threshold", type=int, default=25 * 1024 * 1024)
     p.add_argument("--no-fusion", action="store_true")
     p.add_argument("--exclude-parts", default="")
@@ -86,8 +88,12 @@ def main():
             tuner = ThresholdTuner(opt)
     else:
         from dear_pytorch_amd.parallel import baselines
+        kw = {}
+        if args.method in ("wfbp", "mgwfbp", "asc", "mgs") and \
+                args.compressor != "none" and args.density < 1.0:
+            kw = dict(compressor=args.compressor, density=args.density)
         opt = baselines.make(args.method, base_opt, model,
-                             threshold_bytes=threshold)
+                             threshold_bytes=threshold, **kw)
 
     model.train()
 
