@@ -736,7 +736,7 @@ def test_cnn_batchnorm_dear_ws2_ranks_agree():
         assert torch.isfinite(outs[0][k]).all(), k
 
 
-def _rank_bo_stress(rank, world):
+def _rank_bo_stress(rank, world, pack=False):
     import torch
     import dear_pytorch_amd as dear
     from dear_pytorch_amd.tuner import ThresholdTuner
@@ -748,7 +748,7 @@ def _rank_bo_stress(rank, world):
     dear.broadcast_parameters(m.state_dict(), root_rank=0)
     opt = dear.DistributedOptimizer(
         torch.optim.SGD(m.parameters(), lr=0.01, momentum=0.9), model=m,
-        threshold_bytes=1 << 14)
+        threshold_bytes=1 << 14, pack_grads=pack)
     tuner = ThresholdTuner(opt, bounds_mb=(0.005, 0.5), window=2, warmup=2,
                            trials=8, verbose=False)
     g = torch.Generator().manual_seed(1)
@@ -772,8 +772,9 @@ def _rank_bo_stress(rank, world):
 
 
 @pytest.mark.timeout(600)
-def test_bo_tuning_stress_ws2_stays_consistent():
+@pytest.mark.parametrize("pack", [False, True])
+def test_bo_tuning_stress_ws2_stays_consistent(pack):
     """60 iterations with ~8 live regroups (the riskiest path: buffer
     teardown + hook re-registration mid-training) must keep ranks
-    bit-identical."""
-    assert all(run_dist(_rank_bo_stress, world_size=2))
+    bit-identical — in both grad-view and packed-grad modes."""
+    assert all(run_dist(_rank_bo_stress, world_size=2, args=(pack,)))
