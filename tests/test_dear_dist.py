@@ -529,7 +529,7 @@ def test_bytescheduler_ws2_matches_serial():
         assert torch.equal(outs[0][k], outs[1][k])
 
 
-def _rank_bert_tied(rank, world):
+def _rank_bert_tied(rank, world, pack=False):
     import torch
     import dear_pytorch_amd as dear
     from dear_pytorch_amd import models
@@ -546,7 +546,7 @@ def _rank_bert_tied(rank, world):
     crit = models.BertPretrainingCriterion(cfg.vocab_size)
     opt = dear.DistributedOptimizer(
         torch.optim.SGD(m.parameters(), lr=1e-3), model=m,
-        threshold_bytes=1 << 12)
+        threshold_bytes=1 << 12, pack_grads=pack)
     # tied decoder/embedding weight must occupy exactly one bucket slot
     n_slots = sum(len(g.slots) for g in opt.groups)
     n_params = len({id(p) for p in m.parameters()})
@@ -569,8 +569,9 @@ def _rank_bert_tied(rank, world):
 
 
 @pytest.mark.timeout(300)
-def test_bert_tied_weights_dear_ws2():
-    outs = run_dist(_rank_bert_tied, world_size=2)
+@pytest.mark.parametrize("pack", [False, True])
+def test_bert_tied_weights_dear_ws2(pack):
+    outs = run_dist(_rank_bert_tied, world_size=2, args=(pack,))
     for k in outs[0]:
         assert torch.equal(outs[0][k], outs[1][k]), k
         assert torch.isfinite(outs[0][k]).all(), k
