@@ -81,3 +81,36 @@ def test_chunk_larger_than_credit_still_issues():
     nn.functional.mse_loss(m(x), torch.zeros(2, 64)).backward()
     opt.step()
     assert len(be.issued) >= 1
+
+
+def test_issue_order_deterministic_across_ranks_fuzz():
+    """RCCL safety: two 'ranks' (independent optimizer instances built from
+    identically-seeded models) must issue IDENTICAL collective orders for any
+    model shape / partition / credit combination, because each rank's
+    scheduling decisions depend only on rank-identical state."""
+    import random
+    rng = random.Random(11)
+    for trial in range(10):
+        dims = [rng.choice([4, 8, 16, 32]) for _ in range(rng.randint(2, 6))]
+        part = rng.choice([1024, 2048])  # elements (floor in ctor is 1024)
+        credit = rng.choice([256, 1024, 4096, 1 << 20])
+        orders = []
+        for _rank in range(2):
+            torch.manual_seed(100 + trial)
+            layers = []
+            prev = 8
+            for d in dims:
+                layers.append(nn.Linear(prev, d))
+                prev = d
+            m = nn.Sequential(*layers)
+            be = _RecordingBackend()
+            opt = ByteSchedulerOptimizer(
+                torch.optim.SGD(m.parameters(), lr=0.01), m,
+                partition_bytes=part * 4, credit_bytes=credit, backend=be)
+            x = torch.randn(4, 8, generator=torch.Generator().manual_seed(7))
+            for _ in range(2):
+                nn.functional.mse_loss(m(x), torch.zeros(4, prev)).backward()
+                opt.step()
+            orders.append([_group_of(p, opt.groups) for p in be.issued])
+        assert orders[0] == orders[1], (trial, dims, part, credit)
+        assert len(orders[0]) > 0
