@@ -117,3 +117,34 @@ def test_nearby_layers_grouping():
     groups = build_groups(m, threshold_bytes=None, nearby_layers=2)
     assert len(groups) == 2  # 3 param modules in runs of 2
     assert len(groups[0].modules) == 2 and len(groups[1].modules) == 1
+
+
+def test_padding_aligned_for_all_world_sizes():
+    """Bucket padding invariants for every DP degree we target: padded is a
+    multiple of world*64 elements (256-B-aligned shards per rank) and every
+    slot offset is 64-element aligned."""
+    import torch.nn as nn
+    from dear_pytorch_amd.parallel.fusion import build_groups
+    torch.manual_seed(0)
+    m = nn.Sequential(nn.Linear(17, 33), nn.ReLU(), nn.Linear(33, 7),
+                      nn.Linear(7, 129))  # deliberately odd sizes
+    for world in (1, 2, 4, 8):
+        groups = build_groups(m, threshold_bytes=1 << 10)
+        for g in groups:
+            g.allocate(world, torch.device("cpu"))
+            assert g.padded % (64 * world) == 0, (world, g.padded)
+            assert g.shard.numel() * world == g.padded
+            for s in g.slots:
+                assert s.offset % 64 == 0
+            g.free()
+
+
+def test_summary_and_repr():
+    import torch.nn as nn
+    import dear_pytorch_amd as dear
+    m = nn.Sequential(nn.Linear(8, 8), nn.Linear(8, 4))
+    opt = dear.DistributedOptimizer(torch.optim.SGD(m.parameters(), lr=0.1),
+                                    model=m)
+    s = opt.summary()
+    assert "DeAR plan" in s and "tensors" in s
+    assert "DearOptimizer" in repr(opt)
