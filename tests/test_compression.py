@@ -111,3 +111,48 @@ def test_topk_impl_logic_cpu_stub():
     t = torch.ones(50)
     vals, idx = _topk_abs_impl(t, 7, _TorchKernels)
     assert vals.numel() == 7 and (vals == 1).all()
+
+
+def test_error_feedback_conserves_signal():
+    """Property: for the EF codecs, (sent values) + (stored residual) ==
+    (input + previous residual) — nothing is lost, only deferred."""
+    torch.manual_seed(3)
+    from dear_pytorch_amd.compression import TopKCompressor, EFSignCompressor
+    c = TopKCompressor()
+    prev_resid = torch.zeros(1000)
+    for step in range(4):
+        g = torch.randn(1000)
+        expect_total = g + prev_resid
+        work = g.clone()
+        _, (vals, idx) = c.compress(work, name="p", ratio=0.1)
+        resid = c.residuals["p"]
+        total = resid.clone()
+        total[idx] += vals
+        assert torch.allclose(total, expect_total, atol=1e-6)
+        prev_resid = resid.clone()
+    s = EFSignCompressor()
+    prev = torch.zeros(500)
+    for step in range(3):
+        g = torch.randn(500)
+        expect = g + prev
+        work = g.clone()
+        _, scale = s.compress(work, name="q")
+        # quantized output + residual == accumulated input
+        assert torch.allclose(work.view(-1) + s.residuals["q"], expect,
+                              atol=1e-6)
+        prev = s.residuals["q"].clone()
+
+
+def test_factory_accepts_reference_kwargs():
+    """Horovod-shaped factory signature parity: named_parameters/compression
+    accepted (reference dopt_rsag.py:377-394)."""
+    import torch.nn as nn
+    import dear_pytorch_amd as dear
+    m = nn.Linear(8, 4)
+    opt = dear.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.1), model=m,
+        named_parameters=m.named_parameters(), compression=None)
+    x = torch.randn(2, 8)
+    nn.functional.mse_loss(m(x), torch.zeros(2, 4)).backward()
+    opt.step()
+    opt.synchronize()
